@@ -1,0 +1,230 @@
+#!/usr/bin/env python3
+"""bench.py — driver-contract benchmark of the MI355X Halo2/Pasta backend.
+
+`python bench.py --gpus N --steps K --warmup W [--workload msm|ntt]`
+
+A "step" is one pass of the hot path over one batch of synthetic input.
+Round-1 workload: BASELINE.json configs[1] — the 2^20-point Pallas-family
+(Vesta-curve) variable-base MSM, the dominant stage of create_proof
+(~26 MSM per proof ≈ 40% of prove time). The whole-proof
+workload (configs[3]/[4]) takes over as the default once tg_create_proof
+lands; until then the bench line names this workload in config.workload.
+
+Contract: W untimed warmups, EXACTLY K timed steps bracketed by a
+barrier + torch.cuda.synchronize() on both sides, MAX over ranks, one JSON
+line from rank 0. Inputs (scalars, bases) are resident in HBM before the
+timed region. value = whole-job aggregate across all N GPUs.
+
+Per-rank sharding is embarrassing data parallelism (independent MSMs —
+SURVEY §8e; no collective on the data path), so scaling is "weak".
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+import numpy as np  # noqa: E402
+
+MSM_N = 1 << 20
+NTT_K = 22
+SEED = 0x5441494741  # "TAIGA"
+
+# p (Fp modulus); scalars are clamped below 2^254 < p (uniformity loss is
+# irrelevant for digit statistics)
+P_MOD = 0x40000000000000000000000000000000224698FC094CF91B992D30ED00000001
+
+
+def gen_scalars(n, seed):
+    rng = np.random.Generator(np.random.Philox(seed))
+    arr = rng.integers(0, 2**64, size=(n, 4), dtype=np.uint64)
+    arr[:, 3] &= np.uint64(0x3FFFFFFFFFFFFFFF)  # < 2^254 < p
+    return arr.tobytes()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--workload", choices=["msm", "ntt"], default="msm")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world if world > 1 else args.gpus
+
+    dist = None
+    import torch
+
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="nccl")
+        torch.cuda.set_device(local_rank)
+
+    import taiga_amd
+
+    gpu = taiga_amd.TaigaGpu(local_rank)
+
+    # ---- setup (untimed): synthetic inputs resident in HBM ----
+    if args.workload == "msm":
+        gpu.gen_bases(MSM_N, SEED)  # same base set on every rank
+        scalars = gen_scalars(MSM_N, SEED + 1000 + rank)  # per-rank scalars
+        gpu.scalars_upload(scalars)
+
+        def step():
+            gpu.msm_resident(MSM_N, base_set=0)
+    else:
+        n = 1 << NTT_K
+        poly = gen_scalars(n, SEED + 2000 + rank)
+        gpu.poly_upload(poly, NTT_K)
+
+        def step():
+            gpu.ntt_resident(NTT_K, inverse=False)
+            gpu.ntt_resident(NTT_K, inverse=True)
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        step()
+    gpu.synchronize()
+
+    # profile the timed region's kernels with HIP events on the ctx stream
+    gpu.prof_enable(True)
+    gpu.prof_reset()
+
+    def barrier_sync():
+        if dist:
+            dist.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    gpu.synchronize()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    # ---- roofline (dominant kernel, HIP events on the launch stream) ----
+    if args.workload == "msm":
+        acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
+        # algorithmic bytes per k_bucket_acc launch (BASELINE.md config 2
+        # model): 16 windows x n x (64 B base gather + 4 B sorted-index read)
+        alg_bytes = 16 * MSM_N * 68
+        dom = ("msm_bucket_acc", acc_ms, acc_n, alg_bytes)
+        units_per_step = MSM_N
+        unit = "points/s"
+        metric = "pallas_msm_points_per_sec"
+        workload_name = "msm_2^20_vesta_variable_base (BASELINE configs[1])"
+    else:
+        st_ms, st_n = gpu.prof_get("ntt_fused")
+        s2_ms, s2_n = gpu.prof_get("ntt_stage")
+        # fused kernel: one pass = 2 * n * 32 B (read+write)
+        alg_bytes = 2 * (1 << NTT_K) * 32
+        dom = ("ntt_fused", st_ms, st_n, alg_bytes)
+        units_per_step = 2 * (1 << NTT_K)  # fwd+inv elements
+        unit = "elements/s"
+        metric = "fp_ntt_elements_per_sec"
+        workload_name = "ntt_2^22_fp_roundtrip (BASELINE configs[2])"
+
+    name, tot_ms, cnt, alg_bytes = dom
+    roofline = None
+    if cnt > 0 and tot_ms > 0:
+        avg_s = (tot_ms / cnt) / 1e3
+        achieved = alg_bytes / avg_s / 1e9
+        traffic = os.environ.get("TG_TRAFFIC_BYTES_PER_LAUNCH")
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved, 1),
+            "peak": 8000.0,
+            "unit": "GB/s",
+            "frac": round(achieved / 8000.0, 4),
+            "traffic": float(traffic) if traffic else None,
+            "kernel": name,
+            "avg_launch_ms": round(tot_ms / cnt, 4),
+        }
+
+    value = units_per_step * args.steps * n_gpus / elapsed
+
+    # ---- CPU baseline (oracle "port", rank 0, N=1 only) ----
+    cpu_baseline = None
+    if rank == 0 and world <= 1:
+        sys.path.insert(0, os.path.join(REPO, "oracle"))
+        import oracle_ct as oc
+
+        cores = os.cpu_count()
+        if args.workload == "msm":
+            nb = 1 << 17  # bounded sample (~10-30 s of CPU work)
+            bases = oc.gen_bases(nb, SEED)
+            sc = gen_scalars(nb, SEED + 5000)
+            t0 = time.perf_counter()
+            oc.msm(oc.FQ, sc, bases)
+            dt = time.perf_counter() - t0
+            cpu_baseline = {
+                "value": round(nb / dt, 1),
+                "unit": unit,
+                "cores": cores,
+                "kind": "port",
+                "sample": "one 2^17-point MSM (same scalar distribution), OpenMP",
+            }
+        else:
+            n = 1 << 20  # bounded sample
+            poly = gen_scalars(n, SEED + 6000)
+            t0 = time.perf_counter()
+            f = oc.ntt(oc.FP, 0, 20, poly)
+            oc.ntt(oc.FP, 1, 20, f)
+            dt = time.perf_counter() - t0
+            cpu_baseline = {
+                "value": round(2 * n / dt, 1),
+                "unit": unit,
+                "cores": cores,
+                "kind": "port",
+                "sample": "one 2^20 fwd+inv NTT round-trip, OpenMP",
+            }
+
+    if rank == 0:
+        line = {
+            "metric": metric,
+            "value": round(value, 1),
+            "unit": unit,
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u256",
+            "data": "synthetic",
+            "config": {
+                "workload": workload_name,
+                "n_points": MSM_N if args.workload == "msm" else 1 << NTT_K,
+                "window_bits": 16 if args.workload == "msm" else None,
+                "parallelism": f"dp{n_gpus} (independent MSMs per GPU, no collective)",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line))
+
+    gpu.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,112 @@
+/* taiga_gpu.h — C ABI of the MI355X-native Halo2/Pasta proving backend
+ * (libtaiga_gpu.so).
+ *
+ * This is the drop-in boundary of SURVEY.md §8(b): the entry points are what
+ * a Rust-side binding for Taiga's proof layer would bind, replacing the
+ * internals of:
+ *   Proof::create / Proof::verify      — taiga_halo2/src/proof.rs:25-42,45-54
+ *   Params::<vesta::Affine>::read      — taiga_halo2/src/constant.rs:128-139
+ *   halo2_proofs::plonk::create_proof  — un-vendored heliaxdev/halo2 dep
+ *                                        (call site proof.rs:33)
+ *   best_multiexp / best_fft           — the MSM/NTT engine inside that dep
+ *                                        (microbench entries below)
+ * The Rust-side cbindgen/FFI stub a maintainer would add is shown in
+ * INTEGRATION.md.
+ *
+ * Conventions (identical to the reference wire formats):
+ *   - field elements: 32-byte little-endian canonical repr
+ *     (pasta_curves to_repr(); values >= modulus are rejected)
+ *   - affine points: 64-byte x||y canonical repr; identity = 64 zero bytes
+ *   - compressed points: 32 bytes, sign of y (oddness) in bit 255
+ *     (pasta_curves GroupEncoding; pinned against the bundled SRS)
+ *   - SRS bytes: the exact params_15 file format
+ *     u32(k) ‖ 2^k x 32B g ‖ 2^k x 32B g_lagrange ‖ 32B w ‖ 32B u
+ *
+ * All buffers are caller-owned. A tg_ctx is bound to one GPU and one HIP
+ * stream; use one ctx per device (proofs shard embarrassingly across GPUs —
+ * SURVEY.md §8e, no collective). Calls on one ctx are not thread-safe;
+ * different ctxs are independent.
+ *
+ * Every function returns 0 on success or a negative TG_ERR code.
+ * There is NO CPU fallback anywhere behind this ABI: without a working HIP
+ * device every entry point fails loudly with TG_ERR_HIP.
+ */
+#ifndef TAIGA_GPU_H
+#define TAIGA_GPU_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum {
+  TG_OK = 0,
+  TG_ERR_HIP = -1,       /* HIP runtime failure (incl. no device) */
+  TG_ERR_BADARG = -2,    /* invalid argument / size */
+  TG_ERR_ENCODING = -3,  /* non-canonical field repr or invalid point */
+  TG_ERR_NOSRS = -4,     /* SRS not loaded */
+  TG_ERR_NOMEM = -5,     /* device allocation failed */
+  TG_ERR_STATE = -6,     /* missing resident inputs for a _resident call */
+};
+
+typedef struct tg_ctx tg_ctx;
+
+/* ---- lifecycle ---- */
+int tg_init(int device_ordinal, tg_ctx** out);
+void tg_destroy(tg_ctx* ctx);
+int tg_device_count(void);
+/* last HIP error string for this ctx (valid until next call) */
+const char* tg_error_string(const tg_ctx* ctx);
+
+/* ---- SRS (replaces Params::read + commit bases residency;
+ *      constant.rs:128-139) ----
+ * Parses the params_15 byte format, uploads and decompresses both base sets
+ * into HBM (they stay resident; g_lagrange never changes between proofs). */
+int tg_load_srs(tg_ctx* ctx, const uint8_t* params_bytes, size_t len);
+int tg_srs_k(const tg_ctx* ctx); /* k, or TG_ERR_NOSRS */
+
+/* ---- variable-base MSM over Vesta (microbench entry) ----
+ * base_set: 0 = custom bases (tg_bases_upload), 1 = SRS g,
+ *           2 = SRS g_lagrange.
+ * scalars: n x 32B canonical (vesta::Scalar = pallas::Base = Fp).
+ * out_xy: 64B canonical affine result. */
+int tg_bases_upload(tg_ctx* ctx, const uint8_t* points_xy, size_t n);
+/* synthetic distinct bases generated on-device ([seed+i+1]G) for benches */
+int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed);
+int tg_msm_pallas(tg_ctx* ctx, const uint8_t* scalars, size_t n, int base_set,
+                  uint8_t out_xy[64]);
+
+/* resident-input variant for measurement: scalars staged ahead of the timed
+ * region (bench.py's contract: inputs already in HBM when timing starts). */
+int tg_scalars_upload(tg_ctx* ctx, const uint8_t* scalars, size_t n);
+int tg_msm_resident(tg_ctx* ctx, size_t n, int base_set, uint8_t out_xy[64]);
+
+/* ---- radix-2 NTT over Fp (microbench entry) ----
+ * dir: 0 forward, 1 inverse (inverse includes the n^{-1} scaling).
+ * coset: 0 plain domain (coset variants arrive with the prover pipeline).
+ * poly: in-place 2^k x 32B canonical. */
+int tg_ntt_fp(tg_ctx* ctx, int dir, uint32_t k, int coset, uint8_t* poly);
+
+/* resident-input variant: upload once, run on device data, download. */
+int tg_poly_upload(tg_ctx* ctx, const uint8_t* poly, uint32_t k);
+int tg_ntt_resident(tg_ctx* ctx, int dir, uint32_t k, int coset);
+int tg_poly_download(tg_ctx* ctx, uint8_t* poly, uint32_t k);
+
+/* ---- kernel profiling (HIP events on the ctx stream) ----
+ * names: "msm_digits", "msm_scan", "msm_scatter", "msm_bucket_acc",
+ *        "msm_reduce", "msm_wsum", "ntt_stage", "ntt_fused", "ntt_bitrev",
+ *        "ntt_scale", "msm_total", "ntt_total" */
+void tg_prof_enable(tg_ctx* ctx, int on);
+void tg_prof_reset(tg_ctx* ctx);
+int tg_prof_get(tg_ctx* ctx, const char* name, double* total_ms, long* count);
+
+/* ---- sync ---- */
+int tg_synchronize(tg_ctx* ctx);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* TAIGA_GPU_H */

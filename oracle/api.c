@@ -112,6 +112,32 @@ int orc_pt_on_curve(int fid, const uint8_t* a) {
     return fd_eq(lhs, rhs);
 }
 
+/* generate the bench's synthetic distinct bases: out[i] = [seed+i+1]G on
+ * Vesta, 64-byte canonical affine — same definition as the product's
+ * tg_gen_bases (used by bench.py's cpu_baseline leg and by tests). */
+void orc_gen_bases(long n, uint64_t seed, uint8_t* out) {
+    const fd_ctx* f = &FD_Q;
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+    for (long i = 0; i < n; i++) {
+        pt_aff G;
+        fd_limbs one;
+        fd_one_mont(one, f);
+        fd_neg(G.x, one, f);
+        fd_add(G.y, one, one, f);
+        G.inf = 0;
+        uint64_t k[4] = {seed + (uint64_t)i + 1, 0, 0, 0};
+        pt_jac j, g;
+        pt_from_aff(&g, &G, f);
+        pt_mul(&j, &g, k, f);
+        pt_aff a;
+        pt_to_aff(&a, &j, f);
+        fd_to_bytes(out + 64 * i, a.x, f);
+        fd_to_bytes(out + 64 * i + 32, a.y, f);
+    }
+}
+
 /* --- SRS pin: random-projection check of g_lagrange == group-iNTT(g) ---
  * params: the raw params_15 bytes. Uses `rounds` random projections drawn
  * from a seeded xorshift; full-strength pin of decompression, field, curve,

@@ -115,5 +115,11 @@ def blake2s(data: bytes, personal: bytes = None, outlen: int = 32) -> bytes:
     return out.raw
 
 
+def gen_bases(n: int, seed: int = 0) -> bytes:
+    out = ctypes.create_string_buffer(64 * n)
+    lib().orc_gen_bases(n, ctypes.c_uint64(seed), out)
+    return out.raw
+
+
 def srs_project_check(params: bytes, rounds: int = 2) -> int:
     return lib().orc_srs_project_check(params, len(params), rounds)

@@ -1,0 +1,176 @@
+"""ctypes binding over libtaiga_gpu.so (see include/taiga_gpu.h).
+
+Fails loudly when the extension or a HIP device is missing — no silent
+eager/CPU fallback (the CPU oracle is test-only infrastructure).
+"""
+import ctypes
+import os
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB = os.path.join(_DIR, "csrc", "libtaiga_gpu.so")
+
+ERR_NAMES = {
+    0: "TG_OK",
+    -1: "TG_ERR_HIP",
+    -2: "TG_ERR_BADARG",
+    -3: "TG_ERR_ENCODING",
+    -4: "TG_ERR_NOSRS",
+    -5: "TG_ERR_NOMEM",
+    -6: "TG_ERR_STATE",
+}
+
+
+class TaigaGpuError(RuntimeError):
+    def __init__(self, rc, detail=""):
+        self.rc = rc
+        super().__init__(f"{ERR_NAMES.get(rc, rc)}: {detail}")
+
+
+def lib_path():
+    return _LIB
+
+
+_lib = None
+
+
+def load_library():
+    """Load libtaiga_gpu.so; raises if not built (no fallback)."""
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_LIB):
+            raise TaigaGpuError(
+                -1,
+                f"{_LIB} not built — run `make -C taiga_amd/csrc` "
+                "(hipcc --offload-arch=gfx950); there is no CPU fallback",
+            )
+        lib = ctypes.CDLL(_LIB)
+        lib.tg_init.argtypes = [ctypes.c_int, ctypes.POINTER(ctypes.c_void_p)]
+        lib.tg_error_string.restype = ctypes.c_char_p
+        lib.tg_error_string.argtypes = [ctypes.c_void_p]
+        lib.tg_destroy.argtypes = [ctypes.c_void_p]
+        lib.tg_load_srs.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
+        lib.tg_srs_k.argtypes = [ctypes.c_void_p]
+        lib.tg_bases_upload.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
+        lib.tg_scalars_upload.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
+        lib.tg_gen_bases.argtypes = [ctypes.c_void_p, ctypes.c_size_t, ctypes.c_uint64]
+        lib.tg_msm_pallas.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_int, ctypes.c_char_p,
+        ]
+        lib.tg_msm_resident.argtypes = [
+            ctypes.c_void_p, ctypes.c_size_t, ctypes.c_int, ctypes.c_char_p,
+        ]
+        lib.tg_ntt_fp.argtypes = [
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_uint32, ctypes.c_int, ctypes.c_char_p,
+        ]
+        lib.tg_poly_upload.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint32]
+        lib.tg_ntt_resident.argtypes = [
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_uint32, ctypes.c_int,
+        ]
+        lib.tg_poly_download.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint32]
+        lib.tg_prof_enable.argtypes = [ctypes.c_void_p, ctypes.c_int]
+        lib.tg_prof_enable.restype = None
+        lib.tg_prof_reset.argtypes = [ctypes.c_void_p]
+        lib.tg_prof_reset.restype = None
+        lib.tg_prof_get.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p,
+            ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_long),
+        ]
+        lib.tg_synchronize.argtypes = [ctypes.c_void_p]
+        _lib = lib
+    return _lib
+
+
+def device_count():
+    return load_library().tg_device_count()
+
+
+class TaigaGpu:
+    """One proving context on one GPU (one ctx per device)."""
+
+    def __init__(self, device=0):
+        lib = load_library()
+        h = ctypes.c_void_p()
+        rc = lib.tg_init(device, ctypes.byref(h))
+        if rc != 0:
+            raise TaigaGpuError(rc, lib.tg_error_string(None) or b"")
+        self._lib = lib
+        self._h = h
+
+    def _ck(self, rc):
+        if rc != 0:
+            raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+
+    def close(self):
+        if self._h:
+            self._lib.tg_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    # --- SRS ---
+    def load_srs(self, params_bytes: bytes):
+        self._ck(self._lib.tg_load_srs(self._h, params_bytes, len(params_bytes)))
+
+    @property
+    def srs_k(self):
+        k = self._lib.tg_srs_k(self._h)
+        if k < 0:
+            raise TaigaGpuError(k)
+        return k
+
+    # --- MSM ---
+    def bases_upload(self, points_xy: bytes):
+        self._ck(self._lib.tg_bases_upload(self._h, points_xy, len(points_xy) // 64))
+
+    def scalars_upload(self, scalars: bytes):
+        self._ck(self._lib.tg_scalars_upload(self._h, scalars, len(scalars) // 32))
+
+    def gen_bases(self, n: int, seed: int = 0):
+        self._ck(self._lib.tg_gen_bases(self._h, n, seed))
+
+    def msm(self, scalars: bytes, base_set=0) -> bytes:
+        out = ctypes.create_string_buffer(64)
+        self._ck(self._lib.tg_msm_pallas(self._h, scalars, len(scalars) // 32, base_set, out))
+        return out.raw
+
+    def msm_resident(self, n: int, base_set=0) -> bytes:
+        out = ctypes.create_string_buffer(64)
+        self._ck(self._lib.tg_msm_resident(self._h, n, base_set, out))
+        return out.raw
+
+    # --- NTT ---
+    def ntt(self, data: bytes, k: int, inverse=False, coset=False) -> bytes:
+        buf = ctypes.create_string_buffer(data, len(data))
+        self._ck(self._lib.tg_ntt_fp(self._h, 1 if inverse else 0, k, 1 if coset else 0, buf))
+        return buf.raw
+
+    def poly_upload(self, data: bytes, k: int):
+        self._ck(self._lib.tg_poly_upload(self._h, data, k))
+
+    def ntt_resident(self, k: int, inverse=False, coset=False):
+        self._ck(self._lib.tg_ntt_resident(self._h, 1 if inverse else 0, k, 1 if coset else 0))
+
+    def poly_download(self, k: int) -> bytes:
+        buf = ctypes.create_string_buffer(32 << k)
+        self._ck(self._lib.tg_poly_download(self._h, buf, k))
+        return buf.raw
+
+    # --- profiling ---
+    def prof_enable(self, on=True):
+        self._lib.tg_prof_enable(self._h, 1 if on else 0)
+
+    def prof_reset(self):
+        self._lib.tg_prof_reset(self._h)
+
+    def prof_get(self, name: str):
+        ms = ctypes.c_double()
+        cnt = ctypes.c_long()
+        self._ck(self._lib.tg_prof_get(self._h, name.encode(), ctypes.byref(ms), ctypes.byref(cnt)))
+        return ms.value, cnt.value
+
+    def synchronize(self):
+        self._ck(self._lib.tg_synchronize(self._h))

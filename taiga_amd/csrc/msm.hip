@@ -1,0 +1,316 @@
+// msm.hip — Pippenger variable-base MSM over Vesta for gfx950. PRODUCT CODE.
+//
+// MI355X-native replacement for halo2_proofs' best_multiexp on the
+// create_proof hot path (26+ MSM(2^15) per proof; config 2
+// of BASELINE.json: 2^20 microbench). Parity: vs the CPU oracle, which is
+// pinned to the reference SRS (tests/test_srs_pin.py, test_gpu_parity.py).
+//
+// Plan (all device-side, HBM-bound integer work, no MFMA):
+//   1. k_digits       : 16-bit signed window digits per (point, window),
+//                       histogram via atomics
+//   2. scan           : exclusive prefix sum over the 16x32768 histogram
+//   3. k_scatter      : counting-sort point indices by bucket
+//   4. k_bucket_acc   : one thread per bucket, mixed adds (gathers 64 B
+//                       affine points — the ceil(255/w)*n*68 B bytes model)
+//   5. k_bucket_reduce: per-segment suffix sums -> (V + a*W) partials
+//   6. k_final        : partials -> window sums -> double-and-add combine
+//
+// Scalars arrive in canonical (standard) form — digit extraction needs no
+// Montgomery conversion. Points live device-resident as Montgomery affine
+// 64 B AoS (good 64-byte random-gather granularity).
+
+#include "pasta_device.hpp"
+
+namespace taiga {
+
+constexpr int MSM_C = 16;                    // window bits
+constexpr int MSM_NWIN = 16;                 // ceil(255/16)
+constexpr int MSM_NBUCK = 1 << (MSM_C - 1);  // 32768 buckets (digits 1..32768)
+constexpr int MSM_SEG = 128;                 // buckets per reduction segment
+constexpr int MSM_NSEG = MSM_NBUCK / MSM_SEG;
+
+struct ScalarRepr {
+  u64 l[4];
+};  // canonical LE
+
+// digits: packed u32 = mag(17 bits) | sign<<17 ; mag==0 means skip.
+// hist[w * MSM_NBUCK + (mag-1)]++
+__global__ void k_digits(const ScalarRepr* sc, u64 n, uint32_t* dig, uint32_t* hist) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    ScalarRepr s = sc[i];
+    uint32_t carry = 0;
+#pragma unroll
+    for (int w = 0; w < MSM_NWIN; w++) {
+      int bit0 = w * MSM_C;
+      int limb = bit0 >> 6, sh = bit0 & 63;
+      u64 raw = s.l[limb] >> sh;
+      if (sh && limb < 3) raw |= s.l[limb + 1] << (64 - sh);
+      uint32_t d = (uint32_t)(raw & 0xFFFFu) + carry;
+      uint32_t sign = 0;
+      if (d > (1u << (MSM_C - 1))) {  // d in (2^15, 2^16]: take d - 2^16, carry
+        d = (1u << MSM_C) - d;
+        sign = 1;
+        carry = 1;
+      } else {
+        carry = 0;
+      }
+      uint32_t packed = d ? (d | (sign << 17)) : 0;
+      dig[i * MSM_NWIN + w] = packed;
+      if (d) atomicAdd(&hist[w * MSM_NBUCK + (d - 1)], 1u);
+    }
+    // top window of a <2^255 scalar cannot carry out (bits 240..254 + carry
+    // <= 2^15 fits as a positive digit)
+  }
+}
+
+// ---- 3-kernel exclusive scan over m = MSM_NWIN*MSM_NBUCK entries ----
+__global__ void k_scan_block(const uint32_t* in, uint32_t* out, uint32_t* bsum, u64 m) {
+  __shared__ uint32_t lds[512];
+  u64 base = (u64)blockIdx.x * 512;
+  int t = threadIdx.x;  // 256 threads, 2 elements each
+  lds[t] = base + t < m ? in[base + t] : 0;
+  lds[t + 256] = base + t + 256 < m ? in[base + t + 256] : 0;
+  __syncthreads();
+  // Blelloch up-sweep / down-sweep on 512 elements
+  uint32_t total = 0;
+  for (int off = 1; off < 512; off <<= 1) {
+    int idx = (t + 1) * off * 2 - 1;
+    if (idx < 512) lds[idx] += lds[idx - off];
+    __syncthreads();
+  }
+  if (t == 0) {
+    total = lds[511];
+    lds[511] = 0;
+  }
+  __syncthreads();
+  for (int off = 256; off >= 1; off >>= 1) {
+    int idx = (t + 1) * off * 2 - 1;
+    if (idx < 512) {
+      uint32_t tmp = lds[idx - off];
+      lds[idx - off] = lds[idx];
+      lds[idx] += tmp;
+    }
+    __syncthreads();
+  }
+  if (base + t < m) out[base + t] = lds[t];
+  if (base + t + 256 < m) out[base + t + 256] = lds[t + 256];
+  if (t == 0) bsum[blockIdx.x] = total;
+}
+
+// single-block scan of block sums (nb <= 4096 here: 16*32768/512 = 1024)
+__global__ void k_scan_sums(uint32_t* bsum, u64 nb) {
+  __shared__ uint32_t lds[4096];
+  for (u64 i = threadIdx.x; i < nb; i += blockDim.x) lds[i] = bsum[i];
+  __syncthreads();
+  if (threadIdx.x == 0) {  // serial: nb tiny, once per MSM
+    uint32_t acc = 0;
+    for (u64 i = 0; i < nb; i++) {
+      uint32_t v = lds[i];
+      lds[i] = acc;
+      acc += v;
+    }
+  }
+  __syncthreads();
+  for (u64 i = threadIdx.x; i < nb; i += blockDim.x) bsum[i] = lds[i];
+}
+
+__global__ void k_scan_add(uint32_t* out, const uint32_t* bsum, u64 m) {
+  u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < m) out[i] += bsum[i / 512];
+}
+
+// scatter: sorted[off[bucket]++] = i | sign<<31
+__global__ void k_scatter(const uint32_t* dig, u64 n, uint32_t* off, uint32_t* sorted) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+#pragma unroll
+    for (int w = 0; w < MSM_NWIN; w++) {
+      uint32_t packed = dig[i * MSM_NWIN + w];
+      uint32_t mag = packed & 0x1FFFFu;
+      if (!mag) continue;
+      uint32_t sign = (packed >> 17) & 1u;
+      uint32_t pos = atomicAdd(&off[w * MSM_NBUCK + (mag - 1)], 1u);
+      sorted[pos] = (uint32_t)i | (sign << 31);
+    }
+  }
+}
+
+// bucket accumulation: thread b over all nwin*nbuck buckets
+__global__ void k_bucket_acc(const uint32_t* start, const uint32_t* end,
+                             const uint32_t* sorted, const VestaAff* pts,
+                             VestaJac* buckets, u64 nbuckets_total) {
+  for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < nbuckets_total;
+       b += (u64)gridDim.x * blockDim.x) {
+    uint32_t s = start[b], e = end[b];
+    VestaJac acc = jac_identity<FqCfg>();
+    for (uint32_t idx = s; idx < e; idx++) {
+      uint32_t ent = sorted[idx];
+      VestaAff p = pts[ent & 0x7FFFFFFFu];
+      if (ent >> 31) p = aff_neg(p);
+      acc = jac_add_aff(acc, p);
+    }
+    buckets[b] = acc;
+  }
+}
+
+// segment reduce: for window w, segment g over buckets [g*SEG, (g+1)*SEG):
+// partial = sum_{d in seg} (local_d+1)*B + (g*SEG)*W  where W = sum B.
+__global__ void k_bucket_reduce(const VestaJac* buckets, VestaJac* partials) {
+  u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x;
+  u64 ntot = (u64)MSM_NWIN * MSM_NSEG;
+  for (; t < ntot; t += (u64)gridDim.x * blockDim.x) {
+    u64 w = t / MSM_NSEG;
+    u64 g = t % MSM_NSEG;
+    const VestaJac* B = buckets + w * MSM_NBUCK + g * MSM_SEG;
+    VestaJac run = jac_identity<FqCfg>();
+    VestaJac tot = jac_identity<FqCfg>();
+    for (int d = MSM_SEG - 1; d >= 0; d--) {
+      run = jac_add(run, B[d]);
+      tot = jac_add(tot, run);
+    }
+    // tot = sum (local_d+1) * B ; add base offset: (g*SEG) * run
+    u64 a = (u64)g * MSM_SEG;
+    // double-and-add small scalar a (< 2^15)
+    VestaJac am = jac_identity<FqCfg>();
+    VestaJac base = run;
+    while (a) {
+      if (a & 1) am = jac_add(am, base);
+      base = jac_dbl(base);
+      a >>= 1;
+    }
+    partials[t] = jac_add(tot, am);
+  }
+}
+
+// per-window tree reduce: block w reduces its MSM_NSEG partials to one
+// Jacobian window sum (the 16 window sums go to the host shim, which does
+// the O(1) 240-doubling Horner combine with the same TG_HD primitives —
+// that serial tail is host work, not a 1-lane GPU kernel).
+__global__ void k_wsum(const VestaJac* partials, VestaJac* wsums) {
+  __shared__ VestaJac lds[64];
+  int w = blockIdx.x;
+  int t = threadIdx.x;  // 64 threads: one wave
+  VestaJac acc = jac_identity<FqCfg>();
+  for (int g = t; g < MSM_NSEG; g += 64) acc = jac_add(acc, partials[w * MSM_NSEG + g]);
+  lds[t] = acc;
+  __syncthreads();
+  for (int off = 32; off >= 1; off >>= 1) {
+    if (t < off) lds[t] = jac_add(lds[t], lds[t + off]);
+    __syncthreads();
+  }
+  if (t == 0) wsums[w] = lds[0];
+}
+
+// synthetic bench bases: out[i] = [seed + i + 1] * G (distinct points; the
+// payload values don't affect Pippenger's work profile — digits come from
+// the scalars). Per-thread short double-and-add + one field inversion.
+__global__ void k_gen_bases(VestaAff* out, u64 n, u64 seed) {
+  // generator (-1, 2) in Mont form
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    Fq one = fd_one_mont<FqCfg>();
+    VestaAff G;
+    G.x = fd_neg(one);
+    Fq two = fd_add(one, one);
+    G.y = two;
+    u64 k = seed + i + 1;
+    VestaJac acc = jac_identity<FqCfg>();
+    VestaJac base = jac_from_aff(G);
+    while (k) {
+      if (k & 1) acc = jac_add(acc, base);
+      base = jac_dbl(base);
+      k >>= 1;
+    }
+    out[i] = jac_to_aff(acc);
+  }
+}
+
+// ---- workspace + launcher ----
+struct MsmWork {
+  uint32_t* d_dig = nullptr;     // n * NWIN
+  uint32_t* d_hist = nullptr;    // NWIN * NBUCK   (start offsets after scan)
+  uint32_t* d_off = nullptr;     // running copy for scatter
+  uint32_t* d_end = nullptr;     // end offsets (= start + count)
+  uint32_t* d_bsum = nullptr;    // scan block sums
+  uint32_t* d_sorted = nullptr;  // n * NWIN entries
+  VestaJac* d_buckets = nullptr;
+  VestaJac* d_partials = nullptr;
+  VestaJac* d_wsums = nullptr;  // MSM_NWIN window sums (combined on host)
+  u64 cap_n = 0;
+};
+
+inline hipError_t msm_work_alloc(MsmWork& w, u64 n) {
+  if (w.cap_n >= n) return hipSuccess;
+  u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+  hipError_t e;
+#define TGW_FREE(p) \
+  if (p) { hipFree(p); p = nullptr; }
+  TGW_FREE(w.d_dig) TGW_FREE(w.d_hist) TGW_FREE(w.d_off) TGW_FREE(w.d_end)
+  TGW_FREE(w.d_bsum) TGW_FREE(w.d_sorted) TGW_FREE(w.d_buckets) TGW_FREE(w.d_partials)
+  TGW_FREE(w.d_wsums)
+#undef TGW_FREE
+  if ((e = hipMalloc(&w.d_dig, n * MSM_NWIN * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_hist, m * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_off, m * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_end, m * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_bsum, ((m + 511) / 512) * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_sorted, n * MSM_NWIN * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_buckets, m * sizeof(VestaJac))) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_partials, (u64)MSM_NWIN * MSM_NSEG * sizeof(VestaJac))) !=
+      hipSuccess)
+    return e;
+  if ((e = hipMalloc(&w.d_wsums, MSM_NWIN * sizeof(VestaJac))) != hipSuccess) return e;
+  w.cap_n = n;
+  return hipSuccess;
+}
+
+static inline int msm_grid(u64 work, int block = 256) {
+  u64 blocks = (work + block - 1) / block;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+// scalars: device canonical; pts: device Mont affine; result written to
+// w.d_out as canonical affine x,y (copy back 64 B on host side).
+inline hipError_t msm_run(MsmWork& w, const ScalarRepr* d_scalars, const VestaAff* d_pts,
+                          u64 n, hipStream_t stream) {
+  u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+  hipMemsetAsync(w.d_hist, 0, m * 4, stream);
+  hipLaunchKernelGGL(k_digits, dim3(msm_grid(n)), dim3(256), 0, stream, d_scalars, n,
+                     w.d_dig, w.d_hist);
+  // exclusive scan hist -> start offsets (in d_hist), keep counts via d_end
+  u64 nb = (m + 511) / 512;
+  hipLaunchKernelGGL(k_scan_block, dim3((unsigned)nb), dim3(256), 0, stream, w.d_hist,
+                     w.d_off, w.d_bsum, m);
+  hipLaunchKernelGGL(k_scan_sums, dim3(1), dim3(256), 0, stream, w.d_bsum, nb);
+  hipLaunchKernelGGL(k_scan_add, dim3((unsigned)((m + 255) / 256)), dim3(256), 0, stream,
+                     w.d_off, w.d_bsum, m);
+  // d_off now holds exclusive-scan starts; copy to d_hist (stable starts)
+  hipMemcpyAsync(w.d_hist, w.d_off, m * 4, hipMemcpyDeviceToDevice, stream);
+  hipLaunchKernelGGL(k_scatter, dim3(msm_grid(n)), dim3(256), 0, stream, w.d_dig, n,
+                     w.d_off, w.d_sorted);
+  // after scatter, d_off holds end offsets
+  hipMemcpyAsync(w.d_end, w.d_off, m * 4, hipMemcpyDeviceToDevice, stream);
+  hipLaunchKernelGGL(k_bucket_acc, dim3(msm_grid(m)), dim3(256), 0, stream, w.d_hist,
+                     w.d_end, w.d_sorted, d_pts, w.d_buckets, m);
+  hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)MSM_NWIN * MSM_NSEG)), dim3(256),
+                     0, stream, w.d_buckets, w.d_partials);
+  hipLaunchKernelGGL(k_wsum, dim3(MSM_NWIN), dim3(64), 0, stream, w.d_partials, w.d_wsums);
+  return hipGetLastError();
+}
+
+// host-side final combine: acc = sum_w 2^(16w) * wsum[w]  (Horner, ~240
+// doublings of O(1) work — the same TG_HD primitives as the kernels)
+inline VestaAff msm_host_combine(const VestaJac wsums[MSM_NWIN]) {
+  VestaJac acc = jac_identity<FqCfg>();
+  for (int w = MSM_NWIN - 1; w >= 0; w--) {
+    if (w != MSM_NWIN - 1)
+      for (int b = 0; b < MSM_C; b++) acc = jac_dbl(acc);
+    acc = jac_add(acc, wsums[w]);
+  }
+  return jac_to_aff(acc);
+}
+
+}  // namespace taiga

@@ -1,0 +1,223 @@
+// ntt.hip — radix-2 NTT over Fp for gfx950. PRODUCT CODE.
+//
+// MI355X-native replacement for halo2_proofs' best_fft on the create_proof
+// hot path (SURVEY.md §8a "quotient h(X)" row; the dep is un-vendored —
+// §8c). Semantics (bit-reverse + DIT butterflies, omega from the
+// generator-5 convention) are pinned against the CPU oracle, which is
+// pinned against the reference SRS (tests/test_srs_pin.py).
+//
+// HBM-bound integer work (32 B/element): all kernels are grid-stride with
+// coalesced per-lane 32-byte accesses. v1 = one pass per stage; the fused
+// LDS multi-stage path cuts passes to ceil(k/FUSE) (bytes model in
+// BASELINE.md config 3).
+
+#include "pasta_device.hpp"
+
+namespace taiga {
+
+// tw[i] = omega^i (Mont), i < half_n. log-cost pow per thread.
+template <class C>
+__global__ void k_twiddles(Fd<C>* tw, u64 half_n, int k, bool inverse) {
+  Fd<C> root;
+#pragma unroll
+  for (int i = 0; i < 4; i++) root.l[i] = inverse ? C::ROOT_INV[i] : C::ROOT[i];
+  Fd<C> w = fd_to_mont(root);
+  for (int i = k; i < 32; i++) w = fd_sqr(w);  // omega for 2^k
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < half_n;
+       i += (u64)gridDim.x * blockDim.x) {
+    tw[i] = fd_pow_u64(w, i);
+  }
+}
+
+__device__ __forceinline__ u64 bitrev(u64 x, int k) {
+  u64 r = 0;
+  for (int b = 0; b < k; b++) r |= ((x >> b) & 1ULL) << (k - 1 - b);
+  return r;
+}
+
+// out[i] = to_mont(in[bitrev(i)]); sets *err if any input repr >= MOD.
+template <class C>
+__global__ void k_bitrev_load(Fd<C>* out, const Fd<C>* in, int k, int to_mont_flag,
+                              unsigned* err) {
+  u64 n = 1ULL << k;
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    Fd<C> v = in[bitrev(i, k)];
+    if (err) {
+      // canonical check: v < MOD
+      bool ge = true;  // ge stays true only if v == MOD prefix-equal path
+      bool lt = false;
+#pragma unroll
+      for (int limb = 3; limb >= 0; limb--) {
+        if (!lt && v.l[limb] > C::MOD[limb]) { atomicOr(err, 1u); break; }
+        if (v.l[limb] < C::MOD[limb]) { lt = true; }
+      }
+      if (!lt) atomicOr(err, 1u);
+      (void)ge;
+    }
+    out[i] = to_mont_flag ? fd_to_mont(v) : v;
+  }
+}
+
+// one radix-2 DIT stage s (1-based): butterflies on pairs span 2^(s-1)
+template <class C>
+__global__ void k_ntt_stage(Fd<C>* a, const Fd<C>* tw, int k, int s) {
+  u64 nb = 1ULL << (k - 1);  // number of butterflies
+  u64 half = 1ULL << (s - 1);
+  int tshift = k - s;  // twiddle stride = 2^(k-s)
+  for (u64 j = blockIdx.x * (u64)blockDim.x + threadIdx.x; j < nb;
+       j += (u64)gridDim.x * blockDim.x) {
+    u64 grp = j >> (s - 1);
+    u64 kk = j & (half - 1);
+    u64 i0 = (grp << s) | kk;
+    u64 i1 = i0 + half;
+    Fd<C> t = fd_mul(a[i1], tw[kk << tshift]);
+    Fd<C> lo = a[i0];
+    a[i1] = fd_sub(lo, t);
+    a[i0] = fd_add(lo, t);
+  }
+}
+
+// fused LDS stages: processes FUSE stages [s0+1 .. s0+FUSE] in one pass.
+// Each block handles TILE = 2^FUSE_LOG consecutive butterfly-groups worth of
+// elements gathered with stride 2^s0 elements.
+// Tile size chosen so 256 threads * 2 elements/thread = 512 elements in LDS.
+template <class C, int FUSE>
+__global__ void k_ntt_fused(Fd<C>* a, const Fd<C>* tw, int k, int s0) {
+  // elements per tile
+  constexpr int TILE = 1 << FUSE;           // e.g. 32.. but we use 512 = 2^9
+  __shared__ Fd<C> lds[1 << FUSE];
+  u64 n = 1ULL << k;
+  u64 span = 1ULL << s0;                    // input butterfly span entering this pass
+  // tile t handles elements idx = base + m*span for m in [0, TILE), where
+  // tiles partition each contiguous "segment" of length span*TILE.
+  u64 tiles_per_seg = span;                 // one tile per residue class r < span
+  u64 seg_len = span << FUSE;
+  u64 nseg = n / seg_len;
+  u64 ntiles = nseg * tiles_per_seg;
+  for (u64 tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    u64 seg = tile / tiles_per_seg;
+    u64 r = tile % tiles_per_seg;
+    u64 base = seg * seg_len + r;
+    // load TILE strided elements
+    for (int m = threadIdx.x; m < TILE; m += blockDim.x) lds[m] = a[base + (u64)m * span];
+    __syncthreads();
+    // FUSE local stages; local stage ls corresponds to global stage s0+ls
+    for (int ls = 1; ls <= FUSE; ls++) {
+      int s = s0 + ls;
+      u64 half = 1ULL << (ls - 1);
+      int tshift = k - s;
+      for (int j = threadIdx.x; j < (TILE >> 1); j += blockDim.x) {
+        u64 grp = (u64)j >> (ls - 1);
+        u64 kk = (u64)j & (half - 1);
+        u64 i0 = (grp << ls) | kk;
+        u64 i1 = i0 + half;
+        // global butterfly index kk_global = kk*span + r, twiddle = kk_g << tshift
+        u64 kkg = (kk << s0) | r;
+        Fd<C> t = fd_mul(lds[i1], tw[kkg << tshift]);
+        Fd<C> lo = lds[i0];
+        lds[i1] = fd_sub(lo, t);
+        lds[i0] = fd_add(lo, t);
+      }
+      __syncthreads();
+    }
+    for (int m = threadIdx.x; m < TILE; m += blockDim.x) a[base + (u64)m * span] = lds[m];
+    __syncthreads();
+  }
+}
+
+// pointwise scale (by a constant) and/or from-mont conversion
+template <class C>
+__global__ void k_scale(Fd<C>* a, u64 n, Fd<C> c, int do_scale, int from_mont_flag) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    Fd<C> v = a[i];
+    if (do_scale) v = fd_mul(v, c);
+    if (from_mont_flag) v = fd_from_mont(v);
+    a[i] = v;
+  }
+}
+
+// pointwise multiply by powers of g: a[i] *= g^i  (coset enter/exit)
+template <class C>
+__global__ void k_coset_scale(Fd<C>* a, u64 n, Fd<C> g) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    a[i] = fd_mul(a[i], fd_pow_u64(g, i));
+  }
+}
+
+// ---------- host-side launcher (internal; C ABI wraps it) ----------
+
+struct NttPlan {
+  Fd<FpCfg>* d_tw_fwd = nullptr;  // omega^i, i < n/2
+  Fd<FpCfg>* d_tw_inv = nullptr;
+  int k = -1;
+};
+
+static inline int ntt_grid(u64 work, int block = 256) {
+  u64 blocks = (work + block - 1) / block;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+// builds (or rebuilds) the twiddle tables for size 2^k
+inline hipError_t ntt_plan_init(NttPlan& plan, int k, hipStream_t stream) {
+  if (plan.k == k) return hipSuccess;
+  if (plan.d_tw_fwd) { hipFree(plan.d_tw_fwd); plan.d_tw_fwd = nullptr; }
+  if (plan.d_tw_inv) { hipFree(plan.d_tw_inv); plan.d_tw_inv = nullptr; }
+  u64 half = k > 0 ? (1ULL << (k - 1)) : 1;
+  hipError_t e;
+  if ((e = hipMalloc(&plan.d_tw_fwd, half * sizeof(Fp))) != hipSuccess) return e;
+  if ((e = hipMalloc(&plan.d_tw_inv, half * sizeof(Fp))) != hipSuccess) return e;
+  hipLaunchKernelGGL(k_twiddles<FpCfg>, dim3(ntt_grid(half)), dim3(256), 0, stream,
+                     plan.d_tw_fwd, half, k, false);
+  hipLaunchKernelGGL(k_twiddles<FpCfg>, dim3(ntt_grid(half)), dim3(256), 0, stream,
+                     plan.d_tw_inv, half, k, true);
+  plan.k = k;
+  return hipGetLastError();
+}
+
+// in-place NTT on device Montgomery-form data (d_a), size 2^k.
+// Pre-permuted?  No: performs its own bitrev via d_tmp (ping-pong once).
+// inverse => scales by n^{-1}.
+// FUSE_LOG is the LDS tile log-size for the fused path.
+inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan, int k,
+                          bool inverse, hipStream_t stream, const Fd<FpCfg>* ninv_mont) {
+  const Fd<FpCfg>* tw = inverse ? plan.d_tw_inv : plan.d_tw_fwd;
+  u64 n = 1ULL << k;
+  // bit-reverse into tmp then copy back pointer-swap is managed by caller:
+  // here: tmp <- bitrev(a); stages run in tmp; result copied... to keep the
+  // API simple we bitrev a->tmp, run stages in-place on tmp, then memcpy
+  // back into a (device-to-device, cheap vs stages at v1).
+  hipLaunchKernelGGL(k_bitrev_load<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream, d_tmp,
+                     d_a, k, 0, nullptr);
+  constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB)
+  int s = 1;
+  while (s <= k) {
+    int remaining = k - s + 1;
+    if (remaining >= 2 && k >= FUSE) {
+      int f = remaining < FUSE ? remaining : FUSE;
+      if (f == FUSE) {
+        u64 ntiles = n >> FUSE;
+        hipLaunchKernelGGL((k_ntt_fused<FpCfg, FUSE>),
+                           dim3(ntiles > 2048 ? 2048 : (unsigned)ntiles), dim3(256), 0,
+                           stream, d_tmp, tw, k, s - 1);
+        s += FUSE;
+        continue;
+      }
+    }
+    hipLaunchKernelGGL(k_ntt_stage<FpCfg>, dim3(ntt_grid(n >> 1)), dim3(256), 0, stream,
+                       d_tmp, tw, k, s);
+    s += 1;
+  }
+  if (inverse && ninv_mont) {
+    hipLaunchKernelGGL(k_scale<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream, d_tmp, n,
+                       *ninv_mont, 1, 0);
+  }
+  hipMemcpyAsync(d_a, d_tmp, n * sizeof(Fp), hipMemcpyDeviceToDevice, stream);
+  return hipGetLastError();
+}
+
+}  // namespace taiga

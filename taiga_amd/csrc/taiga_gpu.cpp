@@ -1,0 +1,578 @@
+// taiga_gpu.cpp — C-ABI host shim of libtaiga_gpu.so. PRODUCT CODE.
+//
+// Thin extern "C" layer over the gfx950 kernels (msm.hip, ntt.hip); see
+// include/taiga_gpu.h for the boundary contract and the reference
+// interfaces each entry point replaces. No CPU fallback: every entry point
+// fails with TG_ERR_HIP when no HIP device works.
+
+#include "../../include/taiga_gpu.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "msm.hip"
+#include "ntt.hip"
+
+namespace taiga {
+
+// ---------- small device kernels for API ingestion ----------
+
+// canonical bytes -> Mont, with canonicality check (err bitmask != 0 on bad)
+template <class C>
+__global__ void k_to_mont_check(Fd<C>* out, const Fd<C>* in, u64 n, unsigned* err) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    Fd<C> v = in[i];
+    bool lt = false;
+#pragma unroll
+    for (int limb = 3; limb >= 0; limb--) {
+      if (!lt) {
+        if (v.l[limb] > C::MOD[limb]) { atomicOr(err, 1u); break; }
+        if (v.l[limb] < C::MOD[limb]) lt = true;
+      }
+    }
+    if (!lt) atomicOr(err, 1u);
+    out[i] = fd_to_mont(v);
+  }
+}
+
+template <class C>
+__global__ void k_from_mont(Fd<C>* out, const Fd<C>* in, u64 n) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x)
+    out[i] = fd_from_mont(in[i]);
+}
+
+// canonical affine x||y pairs -> Mont affine (identity = all-zero pair)
+__global__ void k_aff_to_mont_check(VestaAff* out, const VestaAff* in, u64 n,
+                                    unsigned* err) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    VestaAff p = in[i];
+    if (fd_is_zero(p.x) && fd_is_zero(p.y)) {
+      out[i] = p;  // identity stays (0,0)
+      continue;
+    }
+    bool bad = false;
+#pragma unroll
+    for (int c = 0; c < 2; c++) {
+      const Fq& v = c ? p.y : p.x;
+      bool lt = false;
+#pragma unroll
+      for (int limb = 3; limb >= 0; limb--) {
+        if (!lt) {
+          if (v.l[limb] > FqCfg::MOD[limb]) { bad = true; break; }
+          if (v.l[limb] < FqCfg::MOD[limb]) lt = true;
+        }
+      }
+      if (!lt) bad = true;
+    }
+    VestaAff m;
+    m.x = fd_to_mont(p.x);
+    m.y = fd_to_mont(p.y);
+    // on-curve check: y^2 == x^3 + 5
+    Fq five{{5, 0, 0, 0}};
+    Fq rhs = fd_add(fd_mul(fd_sqr(m.x), m.x), fd_to_mont(five));
+    if (!fd_eq(fd_sqr(m.y), rhs)) bad = true;
+    if (bad) atomicOr(err, 1u);
+    out[i] = m;
+  }
+}
+
+// decompress 32-byte compressed points -> Mont affine
+__global__ void k_decompress(VestaAff* out, const uint8_t* in, u64 n, unsigned* err) {
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
+       i += (u64)gridDim.x * blockDim.x) {
+    const uint8_t* b = in + 32 * i;
+    u64 l[4];
+    memcpy(l, b, 32);
+    unsigned sign = (unsigned)(l[3] >> 63);
+    l[3] &= 0x7FFFFFFFFFFFFFFFULL;
+    if ((l[0] | l[1] | l[2] | l[3]) == 0) {
+      if (sign) { atomicOr(err, 1u); }
+      out[i].x = fd_zero<FqCfg>();
+      out[i].y = fd_zero<FqCfg>();
+      continue;
+    }
+    Fq x{{l[0], l[1], l[2], l[3]}};
+    // canonical check
+    bool lt = false, bad = false;
+#pragma unroll
+    for (int limb = 3; limb >= 0; limb--) {
+      if (!lt) {
+        if (x.l[limb] > FqCfg::MOD[limb]) { bad = true; break; }
+        if (x.l[limb] < FqCfg::MOD[limb]) lt = true;
+      }
+    }
+    if (!lt) bad = true;
+    Fq xm = fd_to_mont(x);
+    Fq five{{5, 0, 0, 0}};
+    Fq rhs = fd_add(fd_mul(fd_sqr(xm), xm), fd_to_mont(five));
+    Fq y;
+    if (!bad && !fd_sqrt(y, rhs)) bad = true;
+    if (bad) {
+      atomicOr(err, 1u);
+      out[i].x = fd_zero<FqCfg>();
+      out[i].y = fd_zero<FqCfg>();
+      continue;
+    }
+    if (fd_is_odd_std(y) != (bool)sign) y = fd_neg(y);
+    out[i].x = xm;
+    out[i].y = y;
+  }
+}
+
+// ---------- profiling ----------
+
+struct ProfCounter {
+  std::vector<hipEvent_t> starts, stops;
+  double done_ms = 0;
+  long done_n = 0;
+};
+
+static const char* const PROF_NAMES[] = {
+    "msm_digits", "msm_scan",  "msm_scatter", "msm_bucket_acc", "msm_reduce",
+    "msm_wsum",   "ntt_stage", "ntt_fused",   "ntt_bitrev",     "ntt_scale",
+    "msm_total",  "ntt_total"};
+constexpr int PROF_N = sizeof(PROF_NAMES) / sizeof(PROF_NAMES[0]);
+
+struct Ctx {
+  int device = 0;
+  hipStream_t stream = nullptr;
+  std::string err;
+
+  // SRS
+  int k = -1;
+  VestaAff* d_g = nullptr;
+  VestaAff* d_gl = nullptr;
+  VestaAff h_w, h_u;  // Mont form, host copies
+
+  // custom bases / staged inputs
+  VestaAff* d_bases = nullptr;
+  u64 n_bases = 0;
+  ScalarRepr* d_scalars = nullptr;
+  u64 n_scalars = 0;
+  Fp* d_poly = nullptr;
+  Fp* d_poly_tmp = nullptr;
+  int poly_k = -1;
+
+  MsmWork msm;
+  NttPlan ntt;
+
+  bool prof = false;
+  ProfCounter prof_c[PROF_N];
+};
+
+static thread_local std::string g_err;
+
+static int set_err(Ctx* c, const char* where, hipError_t e) {
+  std::string msg = std::string(where) + ": " + hipGetErrorString(e);
+  if (c)
+    c->err = msg;
+  else
+    g_err = msg;
+  return TG_ERR_HIP;
+}
+
+struct ProfScope {
+  Ctx* c;
+  int idx;
+  ProfScope(Ctx* c_, int idx_) : c(c_), idx(idx_) {
+    if (c->prof) {
+      hipEvent_t ev;
+      hipEventCreate(&ev);
+      hipEventRecord(ev, c->stream);
+      c->prof_c[idx].starts.push_back(ev);
+    }
+  }
+  ~ProfScope() {
+    if (c->prof) {
+      hipEvent_t ev;
+      hipEventCreate(&ev);
+      hipEventRecord(ev, c->stream);
+      c->prof_c[idx].stops.push_back(ev);
+    }
+  }
+};
+
+enum {
+  P_MSM_DIGITS = 0, P_MSM_SCAN, P_MSM_SCATTER, P_MSM_ACC, P_MSM_REDUCE,
+  P_MSM_WSUM, P_NTT_STAGE, P_NTT_FUSED, P_NTT_BITREV, P_NTT_SCALE,
+  P_MSM_TOTAL, P_NTT_TOTAL,
+};
+
+}  // namespace taiga
+
+using namespace taiga;
+
+extern "C" {
+
+int tg_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+const char* tg_error_string(const tg_ctx* ctx) {
+  const Ctx* c = (const Ctx*)ctx;
+  return c ? c->err.c_str() : g_err.c_str();
+}
+
+int tg_init(int device, tg_ctx** out) {
+  if (!out) return TG_ERR_BADARG;
+  hipError_t e = hipSetDevice(device);
+  if (e != hipSuccess) return set_err(nullptr, "hipSetDevice", e);
+  Ctx* c = new Ctx();
+  c->device = device;
+  if ((e = hipStreamCreate(&c->stream)) != hipSuccess) {
+    delete c;
+    return set_err(nullptr, "hipStreamCreate", e);
+  }
+  *out = (tg_ctx*)c;
+  return TG_OK;
+}
+
+void tg_destroy(tg_ctx* ctx) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c) return;
+  hipStreamSynchronize(c->stream);
+#define TGF(p) \
+  if (p) hipFree(p)
+  TGF(c->d_g); TGF(c->d_gl); TGF(c->d_bases); TGF(c->d_scalars);
+  TGF(c->d_poly); TGF(c->d_poly_tmp);
+  TGF(c->msm.d_dig); TGF(c->msm.d_hist); TGF(c->msm.d_off); TGF(c->msm.d_end);
+  TGF(c->msm.d_bsum); TGF(c->msm.d_sorted); TGF(c->msm.d_buckets);
+  TGF(c->msm.d_partials); TGF(c->msm.d_wsums);
+  TGF(c->ntt.d_tw_fwd); TGF(c->ntt.d_tw_inv);
+#undef TGF
+  hipStreamDestroy(c->stream);
+  delete c;
+}
+
+int tg_synchronize(tg_ctx* ctx) {
+  Ctx* c = (Ctx*)ctx;
+  hipError_t e = hipStreamSynchronize(c->stream);
+  if (e != hipSuccess) return set_err(c, "sync", e);
+  return TG_OK;
+}
+
+void tg_prof_enable(tg_ctx* ctx, int on) { ((Ctx*)ctx)->prof = on != 0; }
+
+void tg_prof_reset(tg_ctx* ctx) {
+  Ctx* c = (Ctx*)ctx;
+  for (auto& pc : c->prof_c) {
+    for (auto ev : pc.starts) hipEventDestroy(ev);
+    for (auto ev : pc.stops) hipEventDestroy(ev);
+    pc.starts.clear();
+    pc.stops.clear();
+    pc.done_ms = 0;
+    pc.done_n = 0;
+  }
+}
+
+int tg_prof_get(tg_ctx* ctx, const char* name, double* total_ms, long* count) {
+  Ctx* c = (Ctx*)ctx;
+  hipStreamSynchronize(c->stream);
+  for (int i = 0; i < PROF_N; i++) {
+    if (strcmp(PROF_NAMES[i], name) != 0) continue;
+    ProfCounter& pc = c->prof_c[i];
+    for (size_t j = 0; j < pc.stops.size(); j++) {
+      float ms = 0;
+      hipEventElapsedTime(&ms, pc.starts[j], pc.stops[j]);
+      pc.done_ms += ms;
+      pc.done_n++;
+      hipEventDestroy(pc.starts[j]);
+      hipEventDestroy(pc.stops[j]);
+    }
+    pc.starts.clear();
+    pc.stops.clear();
+    *total_ms = pc.done_ms;
+    *count = pc.done_n;
+    return TG_OK;
+  }
+  return TG_ERR_BADARG;
+}
+
+// ---------- SRS ----------
+
+int tg_load_srs(tg_ctx* ctx, const uint8_t* bytes, size_t len) {
+  Ctx* c = (Ctx*)ctx;
+  if (len < 4) return TG_ERR_BADARG;
+  uint32_t k;
+  memcpy(&k, bytes, 4);
+  if (k > 28) return TG_ERR_BADARG;
+  u64 n = 1ULL << k;
+  if (len != 4 + 2 * n * 32 + 64) return TG_ERR_BADARG;
+  hipError_t e;
+  if (c->d_g) { hipFree(c->d_g); c->d_g = nullptr; }
+  if (c->d_gl) { hipFree(c->d_gl); c->d_gl = nullptr; }
+  if ((e = hipMalloc(&c->d_g, n * sizeof(VestaAff))) != hipSuccess)
+    return set_err(c, "srs alloc g", e);
+  if ((e = hipMalloc(&c->d_gl, n * sizeof(VestaAff))) != hipSuccess)
+    return set_err(c, "srs alloc gl", e);
+  uint8_t* d_comp = nullptr;
+  unsigned* d_err = nullptr;
+  if ((e = hipMalloc(&d_comp, 2 * n * 32 + 64)) != hipSuccess)
+    return set_err(c, "srs alloc comp", e);
+  hipMalloc(&d_err, 4);
+  hipMemsetAsync(d_err, 0, 4, c->stream);
+  hipMemcpyAsync(d_comp, bytes + 4, 2 * n * 32 + 64, hipMemcpyHostToDevice, c->stream);
+  hipLaunchKernelGGL(k_decompress, dim3(msm_grid(n)), dim3(256), 0, c->stream, c->d_g,
+                     d_comp, n, d_err);
+  hipLaunchKernelGGL(k_decompress, dim3(msm_grid(n)), dim3(256), 0, c->stream, c->d_gl,
+                     d_comp + n * 32, n, d_err);
+  // w and u: decompress on device into a scratch, copy to host Mont form
+  VestaAff* d_wu = nullptr;
+  hipMalloc(&d_wu, 2 * sizeof(VestaAff));
+  hipLaunchKernelGGL(k_decompress, dim3(1), dim3(64), 0, c->stream, d_wu,
+                     d_comp + 2 * n * 32, 2, d_err);
+  unsigned h_err = 0;
+  hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, c->stream);
+  VestaAff h_wu[2];
+  hipMemcpyAsync(h_wu, d_wu, 2 * sizeof(VestaAff), hipMemcpyDeviceToHost, c->stream);
+  e = hipStreamSynchronize(c->stream);
+  hipFree(d_comp);
+  hipFree(d_wu);
+  hipFree(d_err);
+  if (e != hipSuccess) return set_err(c, "srs decompress", e);
+  if (h_err) return TG_ERR_ENCODING;
+  c->h_w = h_wu[0];
+  c->h_u = h_wu[1];
+  c->k = (int)k;
+  return TG_OK;
+}
+
+int tg_srs_k(const tg_ctx* ctx) {
+  const Ctx* c = (const Ctx*)ctx;
+  return c->k >= 0 ? c->k : TG_ERR_NOSRS;
+}
+
+// ---------- MSM ----------
+
+int tg_bases_upload(tg_ctx* ctx, const uint8_t* points_xy, size_t n) {
+  Ctx* c = (Ctx*)ctx;
+  if (!n) return TG_ERR_BADARG;
+  hipError_t e;
+  if (c->d_bases && c->n_bases < n) { hipFree(c->d_bases); c->d_bases = nullptr; }
+  if (!c->d_bases &&
+      (e = hipMalloc(&c->d_bases, n * sizeof(VestaAff))) != hipSuccess)
+    return set_err(c, "bases alloc", e);
+  VestaAff* d_raw = nullptr;
+  unsigned* d_err = nullptr;
+  if ((e = hipMalloc(&d_raw, n * sizeof(VestaAff))) != hipSuccess)
+    return set_err(c, "bases raw alloc", e);
+  hipMalloc(&d_err, 4);
+  hipMemsetAsync(d_err, 0, 4, c->stream);
+  hipMemcpyAsync(d_raw, points_xy, n * 64, hipMemcpyHostToDevice, c->stream);
+  hipLaunchKernelGGL(k_aff_to_mont_check, dim3(msm_grid(n)), dim3(256), 0, c->stream,
+                     c->d_bases, d_raw, n, d_err);
+  unsigned h_err = 0;
+  hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, c->stream);
+  e = hipStreamSynchronize(c->stream);
+  hipFree(d_raw);
+  hipFree(d_err);
+  if (e != hipSuccess) return set_err(c, "bases upload", e);
+  if (h_err) return TG_ERR_ENCODING;
+  c->n_bases = n;
+  return TG_OK;
+}
+
+int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed) {
+  Ctx* c = (Ctx*)ctx;
+  if (!n) return TG_ERR_BADARG;
+  hipError_t e;
+  if (c->d_bases && c->n_bases < n) { hipFree(c->d_bases); c->d_bases = nullptr; }
+  if (!c->d_bases &&
+      (e = hipMalloc(&c->d_bases, n * sizeof(VestaAff))) != hipSuccess)
+    return set_err(c, "bases alloc", e);
+  hipLaunchKernelGGL(k_gen_bases, dim3(msm_grid(n)), dim3(256), 0, c->stream, c->d_bases,
+                     n, seed);
+  e = hipStreamSynchronize(c->stream);
+  if (e != hipSuccess) return set_err(c, "gen bases", e);
+  c->n_bases = n;
+  return TG_OK;
+}
+
+int tg_scalars_upload(tg_ctx* ctx, const uint8_t* scalars, size_t n) {
+  Ctx* c = (Ctx*)ctx;
+  if (!n) return TG_ERR_BADARG;
+  hipError_t e;
+  if (c->d_scalars && c->n_scalars < n) { hipFree(c->d_scalars); c->d_scalars = nullptr; }
+  if (!c->d_scalars &&
+      (e = hipMalloc(&c->d_scalars, n * sizeof(ScalarRepr))) != hipSuccess)
+    return set_err(c, "scalars alloc", e);
+  hipMemcpyAsync(c->d_scalars, scalars, n * 32, hipMemcpyHostToDevice, c->stream);
+  e = hipStreamSynchronize(c->stream);
+  if (e != hipSuccess) return set_err(c, "scalars upload", e);
+  c->n_scalars = n;
+  return TG_OK;
+}
+
+static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
+  const VestaAff* bases = nullptr;
+  if (base_set == 0) {
+    if (c->n_bases < n) return TG_ERR_STATE;
+    bases = c->d_bases;
+  } else {
+    if (c->k < 0) return TG_ERR_NOSRS;
+    if (n > (1ULL << c->k)) return TG_ERR_BADARG;
+    bases = base_set == 1 ? c->d_g : c->d_gl;
+  }
+  hipError_t e;
+  if ((e = msm_work_alloc(c->msm, n)) != hipSuccess) return set_err(c, "msm ws", e);
+  {
+    ProfScope total(c, P_MSM_TOTAL);
+    // (inner scopes only bracket sub-phases when profiling)
+    {
+      ProfScope p(c, P_MSM_DIGITS);
+      u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+      hipMemsetAsync(c->msm.d_hist, 0, m * 4, c->stream);
+      hipLaunchKernelGGL(k_digits, dim3(msm_grid(n)), dim3(256), 0, c->stream,
+                         c->d_scalars, n, c->msm.d_dig, c->msm.d_hist);
+    }
+    u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+    u64 nb = (m + 511) / 512;
+    {
+      ProfScope p(c, P_MSM_SCAN);
+      hipLaunchKernelGGL(k_scan_block, dim3((unsigned)nb), dim3(256), 0, c->stream,
+                         c->msm.d_hist, c->msm.d_off, c->msm.d_bsum, m);
+      hipLaunchKernelGGL(k_scan_sums, dim3(1), dim3(256), 0, c->stream, c->msm.d_bsum, nb);
+      hipLaunchKernelGGL(k_scan_add, dim3((unsigned)((m + 255) / 256)), dim3(256), 0,
+                         c->stream, c->msm.d_off, c->msm.d_bsum, m);
+      hipMemcpyAsync(c->msm.d_hist, c->msm.d_off, m * 4, hipMemcpyDeviceToDevice,
+                     c->stream);
+    }
+    {
+      ProfScope p(c, P_MSM_SCATTER);
+      hipLaunchKernelGGL(k_scatter, dim3(msm_grid(n)), dim3(256), 0, c->stream,
+                         c->msm.d_dig, n, c->msm.d_off, c->msm.d_sorted);
+      hipMemcpyAsync(c->msm.d_end, c->msm.d_off, m * 4, hipMemcpyDeviceToDevice,
+                     c->stream);
+    }
+    {
+      ProfScope p(c, P_MSM_ACC);
+      hipLaunchKernelGGL(k_bucket_acc, dim3(msm_grid(m)), dim3(256), 0, c->stream,
+                         c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
+                         c->msm.d_buckets, m);
+    }
+    {
+      ProfScope p(c, P_MSM_REDUCE);
+      hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)MSM_NWIN * MSM_NSEG)),
+                         dim3(256), 0, c->stream, c->msm.d_buckets, c->msm.d_partials);
+    }
+    {
+      ProfScope p(c, P_MSM_WSUM);
+      hipLaunchKernelGGL(k_wsum, dim3(MSM_NWIN), dim3(64), 0, c->stream,
+                         c->msm.d_partials, c->msm.d_wsums);
+    }
+  }
+  VestaJac wsums[MSM_NWIN];
+  hipMemcpyAsync(wsums, c->msm.d_wsums, sizeof(wsums), hipMemcpyDeviceToHost, c->stream);
+  hipError_t es = hipStreamSynchronize(c->stream);
+  if (es != hipSuccess) return set_err(c, "msm run", es);
+  VestaAff r = msm_host_combine(wsums);
+  if (aff_is_identity(r)) {
+    memset(out_xy, 0, 64);
+  } else {
+    Fq x = fd_from_mont(r.x), y = fd_from_mont(r.y);
+    memcpy(out_xy, x.l, 32);
+    memcpy(out_xy + 32, y.l, 32);
+  }
+  return TG_OK;
+}
+
+int tg_msm_resident(tg_ctx* ctx, size_t n, int base_set, uint8_t out_xy[64]) {
+  Ctx* c = (Ctx*)ctx;
+  if (c->n_scalars < n) return TG_ERR_STATE;
+  return msm_common(c, n, base_set, out_xy);
+}
+
+int tg_msm_pallas(tg_ctx* ctx, const uint8_t* scalars, size_t n, int base_set,
+                  uint8_t out_xy[64]) {
+  int rc = tg_scalars_upload(ctx, scalars, n);
+  if (rc != TG_OK) return rc;
+  return msm_common((Ctx*)ctx, n, base_set, out_xy);
+}
+
+// ---------- NTT ----------
+
+int tg_poly_upload(tg_ctx* ctx, const uint8_t* poly, uint32_t k) {
+  Ctx* c = (Ctx*)ctx;
+  if (k > 28) return TG_ERR_BADARG;
+  u64 n = 1ULL << k;
+  hipError_t e;
+  if (c->poly_k != (int)k) {
+    if (c->d_poly) { hipFree(c->d_poly); c->d_poly = nullptr; }
+    if (c->d_poly_tmp) { hipFree(c->d_poly_tmp); c->d_poly_tmp = nullptr; }
+    if ((e = hipMalloc(&c->d_poly, n * sizeof(Fp))) != hipSuccess)
+      return set_err(c, "poly alloc", e);
+    if ((e = hipMalloc(&c->d_poly_tmp, n * sizeof(Fp))) != hipSuccess)
+      return set_err(c, "poly tmp alloc", e);
+    c->poly_k = (int)k;
+  }
+  unsigned* d_err = nullptr;
+  hipMalloc(&d_err, 4);
+  hipMemsetAsync(d_err, 0, 4, c->stream);
+  // upload canonical into tmp, convert to Mont into d_poly
+  hipMemcpyAsync(c->d_poly_tmp, poly, n * 32, hipMemcpyHostToDevice, c->stream);
+  hipLaunchKernelGGL(k_to_mont_check<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, c->stream,
+                     c->d_poly, c->d_poly_tmp, n, d_err);
+  unsigned h_err = 0;
+  hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, c->stream);
+  e = hipStreamSynchronize(c->stream);
+  hipFree(d_err);
+  if (e != hipSuccess) return set_err(c, "poly upload", e);
+  if (h_err) return TG_ERR_ENCODING;
+  return TG_OK;
+}
+
+int tg_ntt_resident(tg_ctx* ctx, int dir, uint32_t k, int coset) {
+  Ctx* c = (Ctx*)ctx;
+  if (coset) return TG_ERR_BADARG;  // arrives with the prover pipeline
+  if (c->poly_k != (int)k) return TG_ERR_STATE;
+  hipError_t e;
+  if ((e = ntt_plan_init(c->ntt, (int)k, c->stream)) != hipSuccess)
+    return set_err(c, "ntt plan", e);
+  Fp ninv;
+  if (dir) {
+    // n^{-1} in Mont form, computed host-side with the shared primitives
+    Fp nstd{{1ULL << k, 0, 0, 0}};
+    ninv = fd_inv(fd_to_mont(nstd));
+  }
+  {
+    ProfScope total(c, P_NTT_TOTAL);
+    if ((e = ntt_run(c->d_poly, c->d_poly_tmp, c->ntt, (int)k, dir != 0, c->stream,
+                     dir ? &ninv : nullptr)) != hipSuccess)
+      return set_err(c, "ntt run", e);
+  }
+  e = hipStreamSynchronize(c->stream);
+  if (e != hipSuccess) return set_err(c, "ntt sync", e);
+  return TG_OK;
+}
+
+int tg_poly_download(tg_ctx* ctx, uint8_t* poly, uint32_t k) {
+  Ctx* c = (Ctx*)ctx;
+  if (c->poly_k != (int)k) return TG_ERR_STATE;
+  u64 n = 1ULL << k;
+  // from_mont into tmp, then D2H
+  hipLaunchKernelGGL(k_from_mont<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, c->stream,
+                     c->d_poly_tmp, c->d_poly, n);
+  hipMemcpyAsync(poly, c->d_poly_tmp, n * 32, hipMemcpyDeviceToHost, c->stream);
+  hipError_t e = hipStreamSynchronize(c->stream);
+  if (e != hipSuccess) return set_err(c, "poly download", e);
+  return TG_OK;
+}
+
+int tg_ntt_fp(tg_ctx* ctx, int dir, uint32_t k, int coset, uint8_t* poly) {
+  int rc = tg_poly_upload(ctx, poly, k);
+  if (rc != TG_OK) return rc;
+  rc = tg_ntt_resident(ctx, dir, k, coset);
+  if (rc != TG_OK) return rc;
+  return tg_poly_download(ctx, poly, k);
+}
+
+}  // extern "C"
