@@ -160,6 +160,18 @@ def main():
 
     value = units_per_step * args.steps * n_gpus / elapsed
 
+    if os.environ.get("TG_PROF_DUMP") and rank == 0:
+        for nm in ["msm_digits", "msm_scan", "msm_scatter", "msm_bucket_acc",
+                   "msm_reduce", "msm_wsum", "msm_total", "ntt_bitrev", "ntt_fused",
+                   "ntt_stage", "ntt_scale", "ntt_total"]:
+            try:
+                ms, cnt = gpu.prof_get(nm)
+                if cnt:
+                    print(f"# prof {nm}: total={ms:.3f} ms n={cnt} avg={ms/cnt:.4f} ms",
+                          file=sys.stderr)
+            except Exception:
+                pass
+
     # ---- CPU baseline (oracle "port", rank 0, N=1 only) ----
     cpu_baseline = None
     if rank == 0 and world <= 1:

@@ -17,7 +17,7 @@ namespace taiga {
 
 // tw[i] = omega^i (Mont), i < half_n. log-cost pow per thread.
 template <class C>
-__global__ void k_twiddles(Fd<C>* tw, u64 half_n, int k, bool inverse) {
+__global__ void __launch_bounds__(256) k_twiddles(Fd<C>* tw, u64 half_n, int k, bool inverse) {
   Fd<C> root;
 #pragma unroll
   for (int i = 0; i < 4; i++) root.l[i] = inverse ? C::ROOT_INV[i] : C::ROOT[i];
@@ -37,7 +37,7 @@ __device__ __forceinline__ u64 bitrev(u64 x, int k) {
 
 // out[i] = to_mont(in[bitrev(i)]); sets *err if any input repr >= MOD.
 template <class C>
-__global__ void k_bitrev_load(Fd<C>* out, const Fd<C>* in, int k, int to_mont_flag,
+__global__ void __launch_bounds__(256) k_bitrev_load(Fd<C>* out, const Fd<C>* in, int k, int to_mont_flag,
                               unsigned* err) {
   u64 n = 1ULL << k;
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
@@ -61,7 +61,7 @@ __global__ void k_bitrev_load(Fd<C>* out, const Fd<C>* in, int k, int to_mont_fl
 
 // one radix-2 DIT stage s (1-based): butterflies on pairs span 2^(s-1)
 template <class C>
-__global__ void k_ntt_stage(Fd<C>* a, const Fd<C>* tw, int k, int s) {
+__global__ void __launch_bounds__(256) k_ntt_stage(Fd<C>* a, const Fd<C>* tw, int k, int s) {
   u64 nb = 1ULL << (k - 1);  // number of butterflies
   u64 half = 1ULL << (s - 1);
   int tshift = k - s;  // twiddle stride = 2^(k-s)
@@ -83,7 +83,7 @@ __global__ void k_ntt_stage(Fd<C>* a, const Fd<C>* tw, int k, int s) {
 // elements gathered with stride 2^s0 elements.
 // Tile size chosen so 256 threads * 2 elements/thread = 512 elements in LDS.
 template <class C, int FUSE>
-__global__ void k_ntt_fused(Fd<C>* a, const Fd<C>* tw, int k, int s0) {
+__global__ void __launch_bounds__(256) k_ntt_fused(Fd<C>* a, const Fd<C>* tw, int k, int s0) {
   // elements per tile
   constexpr int TILE = 1 << FUSE;           // e.g. 32.. but we use 512 = 2^9
   __shared__ Fd<C> lds[1 << FUSE];
@@ -128,7 +128,7 @@ __global__ void k_ntt_fused(Fd<C>* a, const Fd<C>* tw, int k, int s0) {
 
 // pointwise scale (by a constant) and/or from-mont conversion
 template <class C>
-__global__ void k_scale(Fd<C>* a, u64 n, Fd<C> c, int do_scale, int from_mont_flag) {
+__global__ void __launch_bounds__(256) k_scale(Fd<C>* a, u64 n, Fd<C> c, int do_scale, int from_mont_flag) {
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
     Fd<C> v = a[i];
@@ -140,7 +140,7 @@ __global__ void k_scale(Fd<C>* a, u64 n, Fd<C> c, int do_scale, int from_mont_fl
 
 // pointwise multiply by powers of g: a[i] *= g^i  (coset enter/exit)
 template <class C>
-__global__ void k_coset_scale(Fd<C>* a, u64 n, Fd<C> g) {
+__global__ void __launch_bounds__(256) k_coset_scale(Fd<C>* a, u64 n, Fd<C> g) {
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
     a[i] = fd_mul(a[i], fd_pow_u64(g, i));
@@ -179,40 +179,44 @@ inline hipError_t ntt_plan_init(NttPlan& plan, int k, hipStream_t stream) {
   return hipGetLastError();
 }
 
-// in-place NTT on device Montgomery-form data (d_a), size 2^k.
-// Pre-permuted?  No: performs its own bitrev via d_tmp (ping-pong once).
-// inverse => scales by n^{-1}.
-// FUSE_LOG is the LDS tile log-size for the fused path.
+// in-place NTT on device Montgomery-form data (d_a), size 2^k, using d_tmp
+// as the working buffer (bitrev lands there; result copied back).
+// inverse => scales by n^{-1}. ProfFn: callable int->RAII scope, bracketing
+// each launch group for the HIP-event profiler (pass a no-op for none).
+// Prof indices: 0 bitrev, 1 fused, 2 stage, 3 scale.
+template <class ProfFn>
 inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan, int k,
-                          bool inverse, hipStream_t stream, const Fd<FpCfg>* ninv_mont) {
+                          bool inverse, hipStream_t stream, const Fd<FpCfg>* ninv_mont,
+                          ProfFn&& prof) {
   const Fd<FpCfg>* tw = inverse ? plan.d_tw_inv : plan.d_tw_fwd;
   u64 n = 1ULL << k;
-  // bit-reverse into tmp then copy back pointer-swap is managed by caller:
-  // here: tmp <- bitrev(a); stages run in tmp; result copied... to keep the
-  // API simple we bitrev a->tmp, run stages in-place on tmp, then memcpy
-  // back into a (device-to-device, cheap vs stages at v1).
-  hipLaunchKernelGGL(k_bitrev_load<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream, d_tmp,
-                     d_a, k, 0, nullptr);
+  {
+    auto sc = prof(0);
+    hipLaunchKernelGGL(k_bitrev_load<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream,
+                       d_tmp, d_a, k, 0, nullptr);
+  }
   constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB)
   int s = 1;
   while (s <= k) {
     int remaining = k - s + 1;
-    if (remaining >= 2 && k >= FUSE) {
-      int f = remaining < FUSE ? remaining : FUSE;
-      if (f == FUSE) {
-        u64 ntiles = n >> FUSE;
-        hipLaunchKernelGGL((k_ntt_fused<FpCfg, FUSE>),
-                           dim3(ntiles > 2048 ? 2048 : (unsigned)ntiles), dim3(256), 0,
-                           stream, d_tmp, tw, k, s - 1);
-        s += FUSE;
-        continue;
-      }
+    if (remaining >= FUSE && k >= FUSE) {
+      auto sc = prof(1);
+      u64 ntiles = n >> FUSE;
+      hipLaunchKernelGGL((k_ntt_fused<FpCfg, FUSE>),
+                         dim3(ntiles > 2048 ? 2048 : (unsigned)ntiles), dim3(256), 0,
+                         stream, d_tmp, tw, k, s - 1);
+      s += FUSE;
+      continue;
     }
-    hipLaunchKernelGGL(k_ntt_stage<FpCfg>, dim3(ntt_grid(n >> 1)), dim3(256), 0, stream,
-                       d_tmp, tw, k, s);
+    {
+      auto sc = prof(2);
+      hipLaunchKernelGGL(k_ntt_stage<FpCfg>, dim3(ntt_grid(n >> 1)), dim3(256), 0, stream,
+                         d_tmp, tw, k, s);
+    }
     s += 1;
   }
   if (inverse && ninv_mont) {
+    auto sc = prof(3);
     hipLaunchKernelGGL(k_scale<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream, d_tmp, n,
                        *ninv_mont, 1, 0);
   }

@@ -23,7 +23,7 @@ namespace taiga {
 
 // canonical bytes -> Mont, with canonicality check (err bitmask != 0 on bad)
 template <class C>
-__global__ void k_to_mont_check(Fd<C>* out, const Fd<C>* in, u64 n, unsigned* err) {
+__global__ void __launch_bounds__(256) k_to_mont_check(Fd<C>* out, const Fd<C>* in, u64 n, unsigned* err) {
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
     Fd<C> v = in[i];
@@ -41,14 +41,14 @@ __global__ void k_to_mont_check(Fd<C>* out, const Fd<C>* in, u64 n, unsigned* er
 }
 
 template <class C>
-__global__ void k_from_mont(Fd<C>* out, const Fd<C>* in, u64 n) {
+__global__ void __launch_bounds__(256) k_from_mont(Fd<C>* out, const Fd<C>* in, u64 n) {
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x)
     out[i] = fd_from_mont(in[i]);
 }
 
 // canonical affine x||y pairs -> Mont affine (identity = all-zero pair)
-__global__ void k_aff_to_mont_check(VestaAff* out, const VestaAff* in, u64 n,
+__global__ void __launch_bounds__(256) k_aff_to_mont_check(VestaAff* out, const VestaAff* in, u64 n,
                                     unsigned* err) {
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
@@ -84,7 +84,7 @@ __global__ void k_aff_to_mont_check(VestaAff* out, const VestaAff* in, u64 n,
 }
 
 // decompress 32-byte compressed points -> Mont affine
-__global__ void k_decompress(VestaAff* out, const uint8_t* in, u64 n, unsigned* err) {
+__global__ void __launch_bounds__(256) k_decompress(VestaAff* out, const uint8_t* in, u64 n, unsigned* err) {
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
     const uint8_t* b = in + 32 * i;
@@ -466,7 +466,7 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     }
     {
       ProfScope p(c, P_MSM_WSUM);
-      hipLaunchKernelGGL(k_wsum, dim3(MSM_NWIN), dim3(64), 0, c->stream,
+      hipLaunchKernelGGL(k_wsum, dim3(MSM_NWIN), dim3(256), 0, c->stream,
                          c->msm.d_partials, c->msm.d_wsums);
     }
   }
@@ -545,8 +545,11 @@ int tg_ntt_resident(tg_ctx* ctx, int dir, uint32_t k, int coset) {
   }
   {
     ProfScope total(c, P_NTT_TOTAL);
+    static const int NTT_PROF_MAP[4] = {P_NTT_BITREV, P_NTT_FUSED, P_NTT_STAGE,
+                                        P_NTT_SCALE};
+    auto prof = [&](int idx) { return ProfScope(c, NTT_PROF_MAP[idx]); };
     if ((e = ntt_run(c->d_poly, c->d_poly_tmp, c->ntt, (int)k, dir != 0, c->stream,
-                     dir ? &ninv : nullptr)) != hipSuccess)
+                     dir ? &ninv : nullptr, prof)) != hipSuccess)
       return set_err(c, "ntt run", e);
   }
   e = hipStreamSynchronize(c->stream);
