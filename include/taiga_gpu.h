@@ -94,6 +94,24 @@ int tg_poly_upload(tg_ctx* ctx, const uint8_t* poly, uint32_t k);
 int tg_ntt_resident(tg_ctx* ctx, int dir, uint32_t k, int coset);
 int tg_poly_download(tg_ctx* ctx, uint8_t* poly, uint32_t k);
 
+/* ---- proving (replaces plonk::create_proof behind Proof::create;
+ *      proof.rs:25-42 / SURVEY §8a) ----
+ * tg_keygen: parses a TGD1 circuit-description blob (tools/gen_cs1.py) and
+ * builds the proving key on this ctx's GPU (fixed/sigma transforms and
+ * commitments via the NTT/MSM kernels). Requires tg_load_srs first.
+ * tg_create_proof: produces one proof. Round-1 inputs are deterministic
+ * seeds: instance (public input), witness, and prover-randomness streams
+ * (ChaCha20 — BASELINE.md convention; the reference draws from caller RNG,
+ * proof.rs:30). Proof bytes are bit-identical to the CPU oracle on the same
+ * seeds (tests/test_prover_parity.py). */
+int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len);
+int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32],
+                    const uint8_t wit_seed[32], const uint8_t rng_seed[32],
+                    uint8_t* proof_out, size_t cap, size_t* out_len);
+/* blake2b-256 of the generated advice matrix (witness-spec cross-check) */
+int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32],
+                    const uint8_t wit_seed[32], uint8_t out[32]);
+
 /* ---- kernel profiling (HIP events on the ctx stream) ----
  * names: "msm_digits", "msm_scan", "msm_scatter", "msm_bucket_acc",
  *        "msm_reduce", "msm_wsum", "ntt_stage", "ntt_fused", "ntt_bitrev",

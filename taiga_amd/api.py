@@ -67,6 +67,14 @@ def load_library():
             ctypes.c_void_p, ctypes.c_int, ctypes.c_uint32, ctypes.c_int,
         ]
         lib.tg_poly_download.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint32]
+        lib.tg_keygen.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
+        lib.tg_create_proof.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
+            ctypes.c_char_p, ctypes.c_size_t, ctypes.POINTER(ctypes.c_size_t),
+        ]
+        lib.tg_witness_hash.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
+        ]
         lib.tg_prof_enable.argtypes = [ctypes.c_void_p, ctypes.c_int]
         lib.tg_prof_enable.restype = None
         lib.tg_prof_reset.argtypes = [ctypes.c_void_p]
@@ -158,6 +166,22 @@ class TaigaGpu:
         buf = ctypes.create_string_buffer(32 << k)
         self._ck(self._lib.tg_poly_download(self._h, buf, k))
         return buf.raw
+
+    # --- proving ---
+    def keygen(self, desc: bytes):
+        self._ck(self._lib.tg_keygen(self._h, desc, len(desc)))
+
+    def create_proof(self, inst_seed: bytes, wit_seed: bytes, rng_seed: bytes) -> bytes:
+        out = ctypes.create_string_buffer(1 << 16)
+        out_len = ctypes.c_size_t()
+        self._ck(self._lib.tg_create_proof(
+            self._h, inst_seed, wit_seed, rng_seed, out, len(out), ctypes.byref(out_len)))
+        return out.raw[: out_len.value]
+
+    def witness_hash(self, inst_seed: bytes, wit_seed: bytes) -> bytes:
+        out = ctypes.create_string_buffer(32)
+        self._ck(self._lib.tg_witness_hash(self._h, inst_seed, wit_seed, out))
+        return out.raw
 
     # --- profiling ---
     def prof_enable(self, on=True):

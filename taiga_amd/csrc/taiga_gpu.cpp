@@ -16,6 +16,7 @@
 
 #include "msm.hip"
 #include "ntt.hip"
+#include "prover_impl.hpp"
 
 namespace taiga {
 
@@ -140,6 +141,8 @@ static const char* const PROF_NAMES[] = {
     "msm_total",  "ntt_total"};
 constexpr int PROF_N = sizeof(PROF_NAMES) / sizeof(PROF_NAMES[0]);
 
+struct PPk;
+
 struct Ctx {
   int device = 0;
   hipStream_t stream = nullptr;
@@ -162,6 +165,7 @@ struct Ctx {
 
   MsmWork msm;
   NttPlan ntt;
+  struct PPk* ppk = nullptr;  // proving key (prover_gpu.inc)
 
   bool prof = false;
   ProfCounter prof_c[PROF_N];
@@ -207,6 +211,8 @@ enum {
 
 }  // namespace taiga
 
+#include "prover_gpu.inc"
+
 using namespace taiga;
 
 extern "C" {
@@ -240,6 +246,11 @@ void tg_destroy(tg_ctx* ctx) {
   Ctx* c = (Ctx*)ctx;
   if (!c) return;
   hipStreamSynchronize(c->stream);
+  if (c->ppk) {
+    pdev_free(c->ppk->pd);
+    delete c->ppk;
+    c->ppk = nullptr;
+  }
 #define TGF(p) \
   if (p) hipFree(p)
   TGF(c->d_g); TGF(c->d_gl); TGF(c->d_bases); TGF(c->d_scalars);
@@ -579,3 +590,60 @@ int tg_ntt_fp(tg_ctx* ctx, int dir, uint32_t k, int coset, uint8_t* poly) {
 }
 
 }  // extern "C"
+
+/* ---- prover ABI (implementation in prover_gpu.inc) ---- */
+extern "C" {
+
+int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len) {
+  Ctx* c = (Ctx*)ctx;
+  if (c->k < 0) return TG_ERR_NOSRS;
+  if (c->ppk) { delete c->ppk; c->ppk = nullptr; }
+  c->ppk = new PPk();
+  int rc = ppk_keygen(c, *c->ppk, desc, desc_len);
+  if (rc != 0) {
+    delete c->ppk;
+    c->ppk = nullptr;
+  }
+  return rc;
+}
+
+int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
+                    const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
+                    size_t* out_len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  std::vector<uint8_t> proof;
+  int rc = pprove(c, *c->ppk, inst_seed, wit_seed, rng_seed, proof);
+  if (rc != 0) return rc;
+  if (proof.size() > cap) return TG_ERR_BADARG;
+  memcpy(proof_out, proof.data(), proof.size());
+  *out_len = proof.size();
+  return TG_OK;
+}
+
+int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
+                    uint8_t out[32]) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  PDesc& d = c->ppk->d;
+  std::vector<Fp> inst;
+  cs1_instance(d, inst_seed, inst);
+  std::vector<std::vector<Fp>> adv;
+  cs1_witness(d, wit_seed, inst, adv);
+  Blake2b h(32);
+  std::vector<uint8_t> buf(32 * (size_t)d.n);
+  for (int col = 0; col < d.n_advice; col++) {
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+    for (long i = 0; i < d.n; i++) {
+      Fp v = fd_from_mont(adv[col][i]);
+      memcpy(buf.data() + 32 * i, v.l, 32);
+    }
+    h.update(buf.data(), buf.size());
+  }
+  h.final(out);
+  return TG_OK;
+}
+
+}  /* extern "C" */

@@ -1,0 +1,71 @@
+"""Prover parity: the product (GPU) create_proof must be byte-identical to
+the CPU oracle on the same SRS / circuit description / seeds, and the oracle
+verifier must accept the GPU proof (bit-exact parity bar — SURVEY.md §8c)."""
+import ctypes
+import os
+
+import pytest
+
+from conftest import GOLDEN, REPO
+
+pytestmark = pytest.mark.gpu
+
+INST = bytes(32)
+WIT = bytes([1]) + bytes(31)
+RNG = bytes([2]) + bytes(31)
+
+
+@pytest.fixture(scope="module")
+def oracle_pk():
+    lib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+    desc = open(os.path.join(GOLDEN, "cs1.desc"), "rb").read()
+    srs = open(os.path.join(GOLDEN, "params_15"), "rb").read()
+    rc = lib.orc_prover_init(desc, len(desc), srs, len(srs))
+    assert rc in (0, 1)
+    lib.orc_prove_cs1.restype = ctypes.c_long
+    return lib
+
+
+@pytest.fixture(scope="module")
+def gpu_pk(params15):
+    import taiga_amd
+
+    g = taiga_amd.TaigaGpu(0)
+    g.load_srs(params15)
+    desc = open(os.path.join(GOLDEN, "cs1.desc"), "rb").read()
+    g.keygen(desc)
+    yield g
+    g.close()
+
+
+def oracle_prove(lib, inst, wit, rng):
+    out = ctypes.create_string_buffer(1 << 14)
+    n = lib.orc_prove_cs1(inst, wit, rng, out, 1 << 14)
+    assert n > 0, f"oracle prove failed rc={n}"
+    return out.raw[:n]
+
+
+def test_witness_hash_matches(oracle_pk, gpu_pk):
+    got = gpu_pk.witness_hash(INST, WIT)
+    exp = ctypes.create_string_buffer(32)
+    oracle_pk.orc_cs1_witness_hash(INST, WIT, exp)
+    assert got == exp.raw
+
+
+def test_proof_bytes_identical(oracle_pk, gpu_pk):
+    oracle_proof = oracle_prove(oracle_pk, INST, WIT, RNG)
+    gpu_proof = gpu_pk.create_proof(INST, WIT, RNG)
+    assert len(gpu_proof) == len(oracle_proof)
+    assert gpu_proof == oracle_proof
+
+
+def test_gpu_proof_verifies_and_seeds_differ(oracle_pk, gpu_pk):
+    for seed_idx in range(2):
+        inst = bytes([10 + seed_idx]) + bytes(31)
+        wit = bytes([20 + seed_idx]) + bytes(31)
+        rng = bytes([30 + seed_idx]) + bytes(31)
+        p = gpu_pk.create_proof(inst, wit, rng)
+        # oracle verifier accepts the GPU proof
+        assert oracle_pk.orc_verify_cs1(inst, p, len(p)) == 0
+        # and rejects it against a different instance
+        assert oracle_pk.orc_verify_cs1(bytes([99]) + bytes(31), p, len(p)) != 0
