@@ -50,7 +50,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--workload", choices=["msm", "ntt"], default="msm")
+    ap.add_argument("--workload", choices=["proof", "msm", "ntt"], default="proof")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -74,7 +74,25 @@ def main():
     gpu = taiga_amd.TaigaGpu(local_rank)
 
     # ---- setup (untimed): synthetic inputs resident in HBM ----
-    if args.workload == "msm":
+    if args.workload == "proof":
+        # BASELINE configs[3/4] class: Action-circuit-shaped proofs (k=15).
+        # PK + SRS resident on the GPU; witness/instance derived from seeds
+        # per step (fresh randomness each step — nothing cached).
+        import pathlib
+
+        golden = pathlib.Path(REPO) / "tests" / "golden"
+        gpu.load_srs((golden / "params_15").read_bytes())
+        gpu.keygen((golden / "cs1.desc").read_bytes())
+        counter = [0]
+
+        def step():
+            i = counter[0]
+            counter[0] += 1
+            inst = (SEED + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+            wit = (SEED + 77 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+            rng_s = (SEED + 99 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+            gpu.create_proof(inst, wit, rng_s)
+    elif args.workload == "msm":
         gpu.gen_bases(MSM_N, SEED)  # same base set on every rank
         scalars = gen_scalars(MSM_N, SEED + 1000 + rank)  # per-rank scalars
         gpu.scalars_upload(scalars)
@@ -120,7 +138,19 @@ def main():
         elapsed = float(t.item())
 
     # ---- roofline (dominant kernel, HIP events on the launch stream) ----
-    if args.workload == "msm":
+    if args.workload == "proof":
+        acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
+        # per-launch algorithmic bytes at the prover's MSM size (n = 2^15):
+        # 16 windows x n x (64 B base gather + 4 B sorted-index read)
+        alg_bytes = 16 * (1 << 15) * 68
+        dom = ("msm_bucket_acc", acc_ms, acc_n, alg_bytes)
+        units_per_step = 1
+        unit = "proofs/s"
+        metric = "action_proofs_per_sec"
+        workload_name = ("compliance_shaped_proof_k15 (CS1 stand-in: same shape/size as the "
+                         "Action circuit — 10 advice, lookup, 12-col permutation, degree 9; "
+                         "exact compliance witness fidelity is the round-2 item)")
+    elif args.workload == "msm":
         acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
         # algorithmic bytes per k_bucket_acc launch (BASELINE.md config 2
         # model): 16 windows x n x (64 B base gather + 4 B sorted-index read)
@@ -179,7 +209,33 @@ def main():
         import oracle_ct as oc
 
         cores = os.cpu_count()
-        if args.workload == "msm":
+        if args.workload == "proof":
+            lib = oc.lib()
+            import pathlib
+
+            golden = pathlib.Path(REPO) / "tests" / "golden"
+            desc = (golden / "cs1.desc").read_bytes()
+            srs = (golden / "params_15").read_bytes()
+            assert lib.orc_prover_init(desc, len(desc), srs, len(srs)) in (0, 1)
+            import ctypes
+
+            lib.orc_prove_cs1.restype = ctypes.c_long
+            out = ctypes.create_string_buffer(1 << 14)
+            inst = (SEED).to_bytes(32, "little")
+            wit = (SEED + 77).to_bytes(32, "little")
+            rng_s = (SEED + 99).to_bytes(32, "little")
+            t0 = time.perf_counter()
+            nlen = lib.orc_prove_cs1(inst, wit, rng_s, out, 1 << 14)
+            dt = time.perf_counter() - t0
+            assert nlen > 0
+            cpu_baseline = {
+                "value": round(1.0 / dt, 4),
+                "unit": unit,
+                "cores": cores,
+                "kind": "port",
+                "sample": "one CS1 proof via the oracle prover (OpenMP where parallel)",
+            }
+        elif args.workload == "msm":
             nb = 1 << 17  # bounded sample (~10-30 s of CPU work)
             bases = oc.gen_bases(nb, SEED)
             sc = gen_scalars(nb, SEED + 5000)
@@ -224,9 +280,9 @@ def main():
             "data": "synthetic",
             "config": {
                 "workload": workload_name,
-                "n_points": MSM_N if args.workload == "msm" else 1 << NTT_K,
-                "window_bits": 16 if args.workload == "msm" else None,
-                "parallelism": f"dp{n_gpus} (independent MSMs per GPU, no collective)",
+                "n_points": {"proof": 1 << 15, "msm": MSM_N, "ntt": 1 << NTT_K}[args.workload],
+                "window_bits": 16 if args.workload in ("proof", "msm") else None,
+                "parallelism": f"dp{n_gpus} (independent proofs/MSMs per GPU, no collective — SURVEY §8e)",
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
