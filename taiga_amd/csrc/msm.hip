@@ -136,13 +136,24 @@ __global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, uin
   }
 }
 
-// bucket accumulation: thread b over all nwin*nbuck buckets
+// bucket accumulation, phase 1: thread per bucket for small buckets; big
+// buckets (heavily duplicated scalars, e.g. the prover's grand-product tail
+// rows) are deferred to a wave-per-bucket phase-2 kernel — a single lane
+// serially adding tens of thousands of points is a 100x tail otherwise.
+constexpr uint32_t MSM_BIG_BUCKET = 64;
+
 __global__ void __launch_bounds__(256) k_bucket_acc(const uint32_t* start, const uint32_t* end,
                              const uint32_t* sorted, const VestaAff* pts,
-                             VestaJac* buckets, u64 nbuckets_total) {
+                             VestaJac* buckets, u64 nbuckets_total, uint32_t* big_list,
+                             uint32_t* big_count) {
   for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < nbuckets_total;
        b += (u64)gridDim.x * blockDim.x) {
     uint32_t s = start[b], e = end[b];
+    if (e - s > MSM_BIG_BUCKET) {
+      uint32_t slot = atomicAdd(big_count, 1u);
+      big_list[slot] = (uint32_t)b;
+      continue;
+    }
     VestaJac acc = jac_identity<FqCfg>();
     for (uint32_t idx = s; idx < e; idx++) {
       uint32_t ent = sorted[idx];
@@ -151,6 +162,35 @@ __global__ void __launch_bounds__(256) k_bucket_acc(const uint32_t* start, const
       acc = jac_add_aff(acc, p);
     }
     buckets[b] = acc;
+  }
+}
+
+// phase 2: one 64-lane wave per big bucket; lane-strided partials + LDS tree
+__global__ void __launch_bounds__(64) k_bucket_acc_big(const uint32_t* start,
+                                 const uint32_t* end, const uint32_t* sorted,
+                                 const VestaAff* pts, VestaJac* buckets,
+                                 const uint32_t* big_list, const uint32_t* big_count) {
+  __shared__ VestaJac lds[64];
+  uint32_t nbig = *big_count;
+  for (uint32_t gi = blockIdx.x; gi < nbig; gi += gridDim.x) {
+    u64 b = big_list[gi];
+    uint32_t s = start[b], e = end[b];
+    int t = threadIdx.x;
+    VestaJac acc = jac_identity<FqCfg>();
+    for (uint32_t idx = s + t; idx < e; idx += 64) {
+      uint32_t ent = sorted[idx];
+      VestaAff p = pts[ent & 0x7FFFFFFFu];
+      if (ent >> 31) p = aff_neg(p);
+      acc = jac_add_aff(acc, p);
+    }
+    lds[t] = acc;
+    __syncthreads();
+    for (int off = 32; off >= 1; off >>= 1) {
+      if (t < off) lds[t] = jac_add(lds[t], lds[t + off]);
+      __syncthreads();
+    }
+    if (t == 0) buckets[b] = lds[0];
+    __syncthreads();
   }
 }
 
@@ -237,6 +277,7 @@ struct MsmWork {
   VestaJac* d_buckets = nullptr;
   VestaJac* d_partials = nullptr;
   VestaJac* d_wsums = nullptr;  // MSM_NWIN window sums (combined on host)
+  uint32_t* d_big = nullptr;    // big-bucket work list + count (phase 2)
   u64 cap_n = 0;
 };
 
@@ -248,7 +289,7 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n) {
   if (p) { hipFree(p); p = nullptr; }
   TGW_FREE(w.d_dig) TGW_FREE(w.d_hist) TGW_FREE(w.d_off) TGW_FREE(w.d_end)
   TGW_FREE(w.d_bsum) TGW_FREE(w.d_sorted) TGW_FREE(w.d_buckets) TGW_FREE(w.d_partials)
-  TGW_FREE(w.d_wsums)
+  TGW_FREE(w.d_wsums) TGW_FREE(w.d_big)
 #undef TGW_FREE
   if ((e = hipMalloc(&w.d_dig, n * MSM_NWIN * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_hist, m * 4)) != hipSuccess) return e;
@@ -261,6 +302,7 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n) {
       hipSuccess)
     return e;
   if ((e = hipMalloc(&w.d_wsums, MSM_NWIN * sizeof(VestaJac))) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_big, (m + 1) * 4)) != hipSuccess) return e;
   w.cap_n = n;
   return hipSuccess;
 }

@@ -466,9 +466,13 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     }
     {
       ProfScope p(c, P_MSM_ACC);
+      hipMemsetAsync(c->msm.d_big + m, 0, 4, c->stream);
       hipLaunchKernelGGL(k_bucket_acc, dim3(msm_grid(m)), dim3(256), 0, c->stream,
                          c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
-                         c->msm.d_buckets, m);
+                         c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m);
+      hipLaunchKernelGGL(k_bucket_acc_big, dim3(1024), dim3(64), 0, c->stream,
+                         c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
+                         c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
     }
     {
       ProfScope p(c, P_MSM_REDUCE);
