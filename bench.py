@@ -24,6 +24,23 @@ import os
 import sys
 import time
 
+# Cap OpenMP before any library loads: GPU-box containers expose all host
+# cores but enforce a CFS cpu quota — a 256-thread spinning team blows the
+# quota and every parallel region eats ~100 ms throttle stalls.
+def _cpu_quota():
+    try:
+        parts = open("/sys/fs/cgroup/cpu.max").read().split()
+        if parts[0] != "max":
+            return max(1, int(int(parts[0]) / int(parts[1])))
+    except Exception:
+        pass
+    return os.cpu_count() or 8
+
+
+N_CORES = min(_cpu_quota(), os.cpu_count() or 8)
+os.environ.setdefault("OMP_NUM_THREADS", str(N_CORES))
+os.environ.setdefault("OMP_WAIT_POLICY", "PASSIVE")
+
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
@@ -208,7 +225,7 @@ def main():
         sys.path.insert(0, os.path.join(REPO, "oracle"))
         import oracle_ct as oc
 
-        cores = os.cpu_count()
+        cores = N_CORES
         if args.workload == "proof":
             lib = oc.lib()
             import pathlib

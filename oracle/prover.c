@@ -305,7 +305,7 @@ static void poly_eval(fd_limbs out, const fd_limbs* coeff, long n, const fd_limb
     long chunk = (n + T - 1) / T;
     fd_limbs partial[16];
 #ifdef _OPENMP
-#pragma omp parallel for schedule(static) num_threads(T)
+#pragma omp parallel for schedule(static)
 #endif
     for (int t = 0; t < T; t++) {
         long lo = t * chunk, hi = lo + chunk;
@@ -910,7 +910,7 @@ static void inner_prod(fd_limbs out, const fd_limbs* a, const fd_limbs* b, long 
     fd_limbs part[16];
     long chunk = (n + T - 1) / T;
 #ifdef _OPENMP
-#pragma omp parallel for schedule(static) num_threads(T)
+#pragma omp parallel for schedule(static)
 #endif
     for (int t = 0; t < T; t++) {
         long lo = t * chunk, hi = lo + chunk;

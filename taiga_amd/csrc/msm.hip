@@ -142,7 +142,7 @@ __global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, uin
 // serially adding tens of thousands of points is a 100x tail otherwise.
 constexpr uint32_t MSM_BIG_BUCKET = 64;
 
-__global__ void __launch_bounds__(256) k_bucket_acc(const uint32_t* start, const uint32_t* end,
+__global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, const uint32_t* end,
                              const uint32_t* sorted, const VestaAff* pts,
                              VestaJac* buckets, u64 nbuckets_total, uint32_t* big_list,
                              uint32_t* big_count) {
@@ -166,7 +166,7 @@ __global__ void __launch_bounds__(256) k_bucket_acc(const uint32_t* start, const
 }
 
 // phase 2: one 64-lane wave per big bucket; lane-strided partials + LDS tree
-__global__ void __launch_bounds__(64) k_bucket_acc_big(const uint32_t* start,
+__global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
                                  const uint32_t* end, const uint32_t* sorted,
                                  const VestaAff* pts, VestaJac* buckets,
                                  const uint32_t* big_list, const uint32_t* big_count) {
@@ -196,7 +196,7 @@ __global__ void __launch_bounds__(64) k_bucket_acc_big(const uint32_t* start,
 
 // segment reduce: for window w, segment g over buckets [g*SEG, (g+1)*SEG):
 // partial = sum_{d in seg} (local_d+1)*B + (g*SEG)*W  where W = sum B.
-__global__ void __launch_bounds__(256) k_bucket_reduce(const VestaJac* buckets, VestaJac* partials) {
+__global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* buckets, VestaJac* partials) {
   u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x;
   u64 ntot = (u64)MSM_NWIN * MSM_NSEG;
   for (; t < ntot; t += (u64)gridDim.x * blockDim.x) {

@@ -181,7 +181,7 @@ inline Fp ppoly_eval(const Fp* coeff, long n, const Fp& x) {
   long chunk = (n + T - 1) / T;
   Fp partial[T];
 #ifdef _OPENMP
-#pragma omp parallel for schedule(static) num_threads(T)
+#pragma omp parallel for schedule(static)
 #endif
   for (int t = 0; t < T; t++) {
     long lo = t * chunk, hi = std::min(lo + chunk, n);
@@ -199,19 +199,63 @@ inline Fp ppoly_eval(const Fp* coeff, long n, const Fp& x) {
 }
 
 inline void pbatch_inv(Fp* v, long n) {
-  std::vector<Fp> pre(n);
-  Fp run = fd_one_mont<FpCfg>();
-  for (long i = 0; i < n; i++) {
-    pre[i] = run;
-    if (!fd_is_zero(v[i])) run = fd_mul(run, v[i]);
+  // blocked Montgomery trick: one field inversion per block, blocks parallel
+  const int T = 16;
+  long chunk = (n + T - 1) / T;
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (int t = 0; t < T; t++) {
+    long lo = t * chunk, hi = std::min(lo + chunk, n);
+    if (lo >= hi) continue;
+    std::vector<Fp> pre(hi - lo);
+    Fp run = fd_one_mont<FpCfg>();
+    for (long i = lo; i < hi; i++) {
+      pre[i - lo] = run;
+      if (!fd_is_zero(v[i])) run = fd_mul(run, v[i]);
+    }
+    Fp inv = fd_inv(run);
+    for (long i = hi - 1; i >= lo; i--) {
+      if (fd_is_zero(v[i])) continue;
+      Fp tt = fd_mul(inv, pre[i - lo]);
+      inv = fd_mul(inv, v[i]);
+      v[i] = tt;
+    }
   }
-  Fp inv = fd_inv(run);
-  for (long i = n - 1; i >= 0; i--) {
-    if (fd_is_zero(v[i])) continue;
-    Fp t = fd_mul(inv, pre[i]);
-    inv = fd_mul(inv, v[i]);
-    v[i] = t;
+}
+
+// z[0] = init; z[i+1] = z[i] * r[i] for i < u  (parallel blocked scan)
+inline void pprefix_prod(Fp* z, const Fp* r, long u, const Fp& init) {
+  const int T = 16;
+  long chunk = (u + T - 1) / T;
+  Fp bp[T];
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (int t = 0; t < T; t++) {
+    long lo = t * chunk, hi = std::min(lo + chunk, u);
+    Fp p = fd_one_mont<FpCfg>();
+    for (long i = lo; i < hi; i++) p = fd_mul(p, r[i]);
+    bp[t] = p;
   }
+  Fp off[T];
+  Fp acc = init;
+  for (int t = 0; t < T; t++) {
+    off[t] = acc;
+    acc = fd_mul(acc, bp[t]);
+  }
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (int t = 0; t < T; t++) {
+    long lo = t * chunk, hi = std::min(lo + chunk, u);
+    Fp p = off[t];
+    for (long i = lo; i < hi; i++) {
+      z[i] = p;
+      p = fd_mul(p, r[i]);
+    }
+  }
+  z[u] = acc;
 }
 
 inline void pkate_division(Fp* q, const Fp* a, long n, const Fp& b) {
@@ -228,7 +272,7 @@ inline Fp pinner(const Fp* a, const Fp* b, long n) {
   Fp part[T];
   long chunk = (n + T - 1) / T;
 #ifdef _OPENMP
-#pragma omp parallel for schedule(static) num_threads(T)
+#pragma omp parallel for schedule(static)
 #endif
   for (int t = 0; t < T; t++) {
     long lo = t * chunk, hi = std::min(lo + chunk, n);
