@@ -188,22 +188,6 @@ struct Drbg {
   }
 };
 
-// bulk parallel field draws: each draw consumes exactly one 64-byte block,
-// so with rng.pos == 64 the stream is counter-addressable (identical bytes
-// to sequential field() calls)
-inline void drbg_fields_par(Drbg& rng, Fd<FpCfg>* out, long n) {
-  uint32_t base = rng.counter;
-#ifdef _OPENMP
-#pragma omp parallel for schedule(static)
-#endif
-  for (long i = 0; i < n; i++) {
-    uint8_t b[64];
-    Drbg::block(rng.key, base + (uint32_t)i, b);
-    out[i] = from_uniform_512<FpCfg>(b);
-  }
-  rng.counter = base + (uint32_t)n;
-}
-
 // one counter-addressed 64-byte cell (witness generation)
 template <class C>
 inline Fd<C> drbg_cell_field(const uint8_t seed[32], u64 cell) {
@@ -229,6 +213,22 @@ inline Fd<C> from_uniform_512(const uint8_t b[64]) {
 #pragma unroll
   for (int i = 0; i < 4; i++) r2.l[i] = C::R2[i];
   return fd_add(fd_mul(lo, r2), fd_mul(fd_mul(hi, r2), r2));
+}
+
+// bulk parallel field draws: each draw consumes exactly one 64-byte block,
+// so with rng.pos == 64 the stream is counter-addressable (identical bytes
+// to sequential field() calls)
+inline void drbg_fields_par(Drbg& rng, Fd<FpCfg>* out, long n) {
+  uint32_t base = rng.counter;
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (long i = 0; i < n; i++) {
+    uint8_t b[64];
+    Drbg::block(rng.key, base + (uint32_t)i, b);
+    out[i] = from_uniform_512<FpCfg>(b);
+  }
+  rng.counter = base + (uint32_t)n;
 }
 
 // ---------------- transcript (Blake2bWrite/Read over vesta::Affine) ------
