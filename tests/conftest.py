@@ -3,6 +3,22 @@ import sys
 
 import pytest
 
+# Cap OpenMP to the cgroup cpu quota before any library loads (GPU boxes
+# expose 256 cores but enforce a ~16-core quota; oversized spinning teams
+# hit ~100 ms CFS throttle stalls per parallel region).
+def _cpu_quota():
+    try:
+        parts = open("/sys/fs/cgroup/cpu.max").read().split()
+        if parts[0] != "max":
+            return max(1, int(int(parts[0]) / int(parts[1])))
+    except Exception:
+        pass
+    return os.cpu_count() or 8
+
+
+os.environ.setdefault("OMP_NUM_THREADS", str(min(_cpu_quota(), os.cpu_count() or 8)))
+os.environ.setdefault("OMP_WAIT_POLICY", "PASSIVE")
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
 sys.path.insert(0, os.path.join(REPO, "oracle"))
