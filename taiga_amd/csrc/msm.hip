@@ -142,6 +142,7 @@ __global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, uin
 // serially adding tens of thousands of points is a 100x tail otherwise.
 constexpr uint32_t MSM_BIG_BUCKET = 64;
 
+template <bool SAFE>
 __global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, const uint32_t* end,
                              const uint32_t* sorted, const VestaAff* pts,
                              VestaJac* buckets, u64 nbuckets_total, uint32_t* big_list,
@@ -154,18 +155,40 @@ __global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, co
       big_list[slot] = (uint32_t)b;
       continue;
     }
-    VestaJac acc = jac_identity<FqCfg>();
-    for (uint32_t idx = s; idx < e; idx++) {
-      uint32_t ent = sorted[idx];
-      VestaAff p = pts[ent & 0x7FFFFFFFu];
-      if (ent >> 31) p = aff_neg(p);
-      acc = jac_add_aff(acc, p);
+    if (!SAFE) {
+      if (s == e) {
+        buckets[b] = jac_identity<FqCfg>();
+        continue;
+      }
+      uint32_t ent0 = sorted[s];
+      VestaAff p0 = pts[ent0 & 0x7FFFFFFFu];
+      if (ent0 >> 31) p0 = aff_neg_fast(p0);
+      VestaJac acc;
+      acc.x = p0.x;
+      acc.y = p0.y;
+      acc.z = fd_one_mont<FqCfg>();
+      for (uint32_t idx = s + 1; idx < e; idx++) {
+        uint32_t ent = sorted[idx];
+        VestaAff p = pts[ent & 0x7FFFFFFFu];
+        if (ent >> 31) p = aff_neg_fast(p);
+        jac_add_aff_fast(acc, p);
+      }
+      buckets[b] = acc;
+    } else {
+      VestaJac acc = jac_identity<FqCfg>();
+      for (uint32_t idx = s; idx < e; idx++) {
+        uint32_t ent = sorted[idx];
+        VestaAff p = pts[ent & 0x7FFFFFFFu];
+        if (ent >> 31) p = aff_neg(p);
+        jac_add_aff_inplace(acc, p);
+      }
+      buckets[b] = acc;
     }
-    buckets[b] = acc;
   }
 }
 
 // phase 2: one 64-lane wave per big bucket; lane-strided partials + LDS tree
+template <bool SAFE>
 __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
                                  const uint32_t* end, const uint32_t* sorted,
                                  const VestaAff* pts, VestaJac* buckets,
@@ -177,11 +200,29 @@ __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
     uint32_t s = start[b], e = end[b];
     int t = threadIdx.x;
     VestaJac acc = jac_identity<FqCfg>();
-    for (uint32_t idx = s + t; idx < e; idx += 64) {
-      uint32_t ent = sorted[idx];
-      VestaAff p = pts[ent & 0x7FFFFFFFu];
-      if (ent >> 31) p = aff_neg(p);
-      acc = jac_add_aff(acc, p);
+    if (!SAFE) {
+      uint32_t first = s + (uint32_t)t;
+      if (first < e) {
+        uint32_t ent0 = sorted[first];
+        VestaAff p0 = pts[ent0 & 0x7FFFFFFFu];
+        if (ent0 >> 31) p0 = aff_neg_fast(p0);
+        acc.x = p0.x;
+        acc.y = p0.y;
+        acc.z = fd_one_mont<FqCfg>();
+        for (uint32_t idx = first + 64; idx < e; idx += 64) {
+          uint32_t ent = sorted[idx];
+          VestaAff p = pts[ent & 0x7FFFFFFFu];
+          if (ent >> 31) p = aff_neg_fast(p);
+          jac_add_aff_fast(acc, p);
+        }
+      }
+    } else {
+      for (uint32_t idx = s + t; idx < e; idx += 64) {
+        uint32_t ent = sorted[idx];
+        VestaAff p = pts[ent & 0x7FFFFFFFu];
+        if (ent >> 31) p = aff_neg(p);
+        jac_add_aff_inplace(acc, p);
+      }
     }
     lds[t] = acc;
     __syncthreads();

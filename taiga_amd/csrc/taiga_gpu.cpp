@@ -424,6 +424,9 @@ int tg_scalars_upload(tg_ctx* ctx, const uint8_t* scalars, size_t n) {
 }
 
 static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
+  // custom uploaded bases may contain identities/duplicates -> SAFE variant;
+  // SRS bases are distinct non-identity points -> fast branchless variant
+  const bool safe = base_set == 0;
   const VestaAff* bases = nullptr;
   if (base_set == 0) {
     if (c->n_bases < n) return TG_ERR_STATE;
@@ -467,12 +470,21 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     {
       ProfScope p(c, P_MSM_ACC);
       hipMemsetAsync(c->msm.d_big + m, 0, 4, c->stream);
-      hipLaunchKernelGGL(k_bucket_acc, dim3(msm_grid(m)), dim3(256), 0, c->stream,
-                         c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
-                         c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m);
-      hipLaunchKernelGGL(k_bucket_acc_big, dim3(1024), dim3(64), 0, c->stream,
-                         c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
-                         c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
+      if (safe) {
+        hipLaunchKernelGGL(k_bucket_acc<true>, dim3(msm_grid(m)), dim3(256), 0, c->stream,
+                           c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
+                           c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m);
+        hipLaunchKernelGGL(k_bucket_acc_big<true>, dim3(1024), dim3(64), 0, c->stream,
+                           c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
+                           c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
+      } else {
+        hipLaunchKernelGGL(k_bucket_acc<false>, dim3(msm_grid(m)), dim3(256), 0, c->stream,
+                           c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
+                           c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m);
+        hipLaunchKernelGGL(k_bucket_acc_big<false>, dim3(1024), dim3(64), 0, c->stream,
+                           c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
+                           c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
+      }
     }
     {
       ProfScope p(c, P_MSM_REDUCE);
