@@ -157,6 +157,7 @@ struct Ctx {
   // custom bases / staged inputs
   VestaAff* d_bases = nullptr;
   u64 n_bases = 0;
+  bool bases_distinct = false;  // gen_bases output: distinct, non-identity
   ScalarRepr* d_scalars = nullptr;
   u64 n_scalars = 0;
   Fp* d_poly = nullptr;
@@ -389,6 +390,7 @@ int tg_bases_upload(tg_ctx* ctx, const uint8_t* points_xy, size_t n) {
   if (e != hipSuccess) return set_err(c, "bases upload", e);
   if (h_err) return TG_ERR_ENCODING;
   c->n_bases = n;
+  c->bases_distinct = false;
   return TG_OK;
 }
 
@@ -405,6 +407,7 @@ int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed) {
   e = hipStreamSynchronize(c->stream);
   if (e != hipSuccess) return set_err(c, "gen bases", e);
   c->n_bases = n;
+  c->bases_distinct = true;
   return TG_OK;
 }
 
@@ -426,7 +429,7 @@ int tg_scalars_upload(tg_ctx* ctx, const uint8_t* scalars, size_t n) {
 static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
   // custom uploaded bases may contain identities/duplicates -> SAFE variant;
   // SRS bases are distinct non-identity points -> fast branchless variant
-  const bool safe = base_set == 0;
+  const bool safe = base_set == 0 && !c->bases_distinct;
   const VestaAff* bases = nullptr;
   if (base_set == 0) {
     if (c->n_bases < n) return TG_ERR_STATE;
