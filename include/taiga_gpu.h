@@ -108,6 +108,10 @@ int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len);
 int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], const uint8_t rng_seed[32],
                     uint8_t* proof_out, size_t cap, size_t* out_len);
+/* verify one proof (replaces Proof::verify / plonk::verify_proof with the
+ * SingleVerifier strategy — proof.rs:45-54). Returns TG_OK iff valid. */
+int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* proof,
+                    size_t proof_len);
 /* blake2b-256 of the generated advice matrix (witness-spec cross-check) */
 int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], uint8_t out[32]);

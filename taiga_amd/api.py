@@ -72,6 +72,9 @@ def load_library():
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
             ctypes.c_char_p, ctypes.c_size_t, ctypes.POINTER(ctypes.c_size_t),
         ]
+        lib.tg_verify_proof.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
+        ]
         lib.tg_witness_hash.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
         ]
@@ -177,6 +180,14 @@ class TaigaGpu:
         self._ck(self._lib.tg_create_proof(
             self._h, inst_seed, wit_seed, rng_seed, out, len(out), ctypes.byref(out_len)))
         return out.raw[: out_len.value]
+
+    def verify_proof(self, inst_seed: bytes, proof: bytes) -> bool:
+        rc = self._lib.tg_verify_proof(self._h, inst_seed, proof, len(proof))
+        if rc == 0:
+            return True
+        if rc == -1:
+            return False
+        raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
 
     def witness_hash(self, inst_seed: bytes, wit_seed: bytes) -> bytes:
         out = ctypes.create_string_buffer(32)

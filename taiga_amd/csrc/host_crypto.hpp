@@ -296,6 +296,59 @@ struct Transcript {
     clone.final(dig);
     return from_uniform_512<FpCfg>(dig);
   }
+
+  // ---- read mode (verifier) ----
+  static bool fq_canonical(const Fq& v) {
+    for (int limb = 3; limb >= 0; limb--) {
+      if (v.l[limb] > FqCfg::MOD[limb]) return false;
+      if (v.l[limb] < FqCfg::MOD[limb]) return true;
+    }
+    return false;
+  }
+
+  // 32-byte compressed -> Mont affine (host Tonelli-Shanks); rejects
+  // identity and invalid encodings (transcript points are never identity)
+  static bool decompress(VestaAff& out, const uint8_t in[32]) {
+    u64 l[4];
+    memcpy(l, in, 32);
+    unsigned sign = (unsigned)(l[3] >> 63);
+    l[3] &= 0x7FFFFFFFFFFFFFFFULL;
+    if ((l[0] | l[1] | l[2] | l[3]) == 0) return false;
+    Fq x{{l[0], l[1], l[2], l[3]}};
+    if (!fq_canonical(x)) return false;
+    Fq xm = fd_to_mont(x);
+    Fq five{{5, 0, 0, 0}};
+    Fq rhs = fd_add(fd_mul(fd_sqr(xm), xm), fd_to_mont(five));
+    Fq y;
+    if (!fd_sqrt(y, rhs)) return false;
+    if (fd_is_odd_std(y) != (bool)sign) y = fd_neg(y);
+    out.x = xm;
+    out.y = y;
+    return true;
+  }
+
+  bool read_point(VestaAff& p) {
+    if (rpos + 32 > rlen) return false;
+    if (!decompress(p, rbuf + rpos)) return false;
+    rpos += 32;
+    common_point(p);
+    return true;
+  }
+
+  bool read_scalar(Fp& s) {
+    if (rpos + 32 > rlen) return false;
+    Fp v;
+    memcpy(v.l, rbuf + rpos, 32);
+    for (int limb = 3;; limb--) {
+      if (v.l[limb] > FpCfg::MOD[limb]) return false;
+      if (v.l[limb] < FpCfg::MOD[limb]) break;
+      if (limb == 0) return false;
+    }
+    rpos += 32;
+    s = fd_to_mont(v);
+    common_scalar(s);
+    return true;
+  }
 };
 
 }  // namespace taiga

@@ -666,3 +666,14 @@ int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_
 }
 
 }  /* extern "C" */
+
+extern "C" {
+
+int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* proof,
+                    size_t proof_len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  return pverify(c, *c->ppk, inst_seed, proof, proof_len);
+}
+
+}  /* extern "C" */

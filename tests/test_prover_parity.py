@@ -59,6 +59,19 @@ def test_proof_bytes_identical(oracle_pk, gpu_pk):
     assert gpu_proof == oracle_proof
 
 
+def test_product_verifier(oracle_pk, gpu_pk):
+    """tg_verify_proof accepts oracle and GPU proofs, rejects tampering."""
+    oracle_proof = oracle_prove(oracle_pk, INST, WIT, RNG)
+    assert gpu_pk.verify_proof(INST, oracle_proof)
+    gpu_proof = gpu_pk.create_proof(INST, WIT, RNG)
+    assert gpu_pk.verify_proof(INST, gpu_proof)
+    bad = bytearray(gpu_proof)
+    bad[50] ^= 1
+    assert not gpu_pk.verify_proof(INST, bytes(bad))
+    assert not gpu_pk.verify_proof(bytes([9]) + bytes(31), gpu_proof)
+    assert not gpu_pk.verify_proof(INST, gpu_proof[:-32])
+
+
 def test_gpu_proof_verifies_and_seeds_differ(oracle_pk, gpu_pk):
     for seed_idx in range(2):
         inst = bytes([10 + seed_idx]) + bytes(31)
