@@ -185,7 +185,8 @@ class TaigaGpu:
         rc = self._lib.tg_verify_proof(self._h, inst_seed, proof, len(proof))
         if rc == 0:
             return True
-        if rc == -1:
+        # -1 = final check failed; -1xx = malformed/truncated transcript
+        if rc == -1 or rc <= -100:
             return False
         raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
 
