@@ -285,6 +285,35 @@ inline Fp pinner(const Fp* a, const Fp* b, long n) {
   return out;
 }
 
+// fixed-point window table: [j] holds tab[w][d] = [d * 2^(8w)] P for
+// d in 1..255, w in 0..31 — turns a 255-bit scalar mult of a FIXED point
+// (W, U) into <=32 mixed adds.
+struct FixedMulTab {
+  std::vector<VestaJac> tab;  // 32 windows * 255 digits (jacobian: no
+                              // per-entry inversion at init)
+  void init(const VestaAff& P) {
+    tab.resize(32 * 255);
+    VestaJac base = jac_from_aff(P);
+    for (int w = 0; w < 32; w++) {
+      VestaJac acc = base;
+      for (int dd = 1; dd <= 255; dd++) {
+        tab[w * 255 + (dd - 1)] = acc;
+        acc = jac_add(acc, base);
+      }
+      base = acc;  // = 256 * base = [2^(8(w+1))] P
+    }
+  }
+  VestaJac mul(const Fd<FpCfg>& s_mont) const {
+    Fd<FpCfg> s = fd_from_mont(s_mont);
+    VestaJac acc = jac_identity<FqCfg>();
+    for (int w = 0; w < 32; w++) {
+      unsigned dd = (unsigned)((s.l[w / 8] >> (8 * (w % 8))) & 0xFF);
+      if (dd) acc = jac_add(acc, tab[w * 255 + (dd - 1)]);
+    }
+    return acc;
+  }
+};
+
 // host jacobian helpers (host side of MSM combine)
 inline VestaJac jac_mul_host(const VestaJac& p, const Fp& s_mont) {
   Fp s = fd_from_mont(s_mont);
