@@ -35,9 +35,13 @@ struct ScalarRepr {
 
 // digits: packed u32 = mag(17 bits) | sign<<17 ; mag==0 means skip.
 // hist[w * MSM_NBUCK + (mag-1)]++
-__global__ void __launch_bounds__(256) k_digits(const ScalarRepr* sc, u64 n, uint32_t* dig, uint32_t* hist) {
+__global__ void __launch_bounds__(256) k_digits(const ScalarRepr* sc, u64 n, u64 nn, uint32_t* dig,
+                uint32_t* hist) {
+  // n = total scalars (= nn * batch); histogram is per batch b = i / nn
+  const u64 m = (u64)MSM_NWIN * MSM_NBUCK;
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
+    u64 bofs = (i / nn) * m;
     ScalarRepr s = sc[i];
     uint32_t carry = 0;
 #pragma unroll
@@ -57,7 +61,7 @@ __global__ void __launch_bounds__(256) k_digits(const ScalarRepr* sc, u64 n, uin
       }
       uint32_t packed = d ? (d | (sign << 17)) : 0;
       dig[i * MSM_NWIN + w] = packed;
-      if (d) atomicAdd(&hist[w * MSM_NBUCK + (d - 1)], 1u);
+      if (d) atomicAdd(&hist[bofs + w * MSM_NBUCK + (d - 1)], 1u);
     }
     // top window of a <2^255 scalar cannot carry out (bits 240..254 + carry
     // <= 2^15 fits as a positive digit)
@@ -120,18 +124,51 @@ __global__ void __launch_bounds__(256) k_scan_add(uint32_t* out, const uint32_t*
   if (i < m) out[i] += bsum[i / 512];
 }
 
+// exclusive scan over m entries (two-level when m/512 exceeds one block's
+// LDS); bsum workspace holds level-1 (+ level-2) block sums.
+static inline int msm_grid_(u64 work, int block = 256) {
+  u64 blocks = (work + block - 1) / block;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+inline void msm_scan(const uint32_t* in, uint32_t* out, uint32_t* bsum, u64 m,
+                     hipStream_t stream) {
+  u64 nb1 = (m + 511) / 512;
+  hipLaunchKernelGGL(k_scan_block, dim3((unsigned)nb1), dim3(256), 0, stream, in, out,
+                     bsum, m);
+  if (nb1 <= 4096) {
+    hipLaunchKernelGGL(k_scan_sums, dim3(1), dim3(256), 0, stream, bsum, nb1);
+  } else {
+    u64 nb2 = (nb1 + 511) / 512;
+    uint32_t* bsum2 = bsum + nb1;
+    hipLaunchKernelGGL(k_scan_block, dim3((unsigned)nb2), dim3(256), 0, stream, bsum,
+                       bsum, bsum2, nb1);
+    hipLaunchKernelGGL(k_scan_sums, dim3(1), dim3(256), 0, stream, bsum2, nb2);
+    hipLaunchKernelGGL(k_scan_add, dim3((unsigned)((nb1 + 255) / 256)), dim3(256), 0,
+                       stream, bsum, bsum2, nb1);
+  }
+  hipLaunchKernelGGL(k_scan_add, dim3((unsigned)((m + 255) / 256)), dim3(256), 0, stream,
+                     out, bsum, m);
+}
+
 // scatter: sorted[off[bucket]++] = i | sign<<31
-__global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, uint32_t* off, uint32_t* sorted) {
+__global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, u64 nn, uint32_t* off,
+                 uint32_t* sorted) {
+  const u64 m = (u64)MSM_NWIN * MSM_NBUCK;
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
+    u64 bofs = (i / nn) * m;
+    uint32_t il = (uint32_t)(i % nn);  // batch-local point index
 #pragma unroll
     for (int w = 0; w < MSM_NWIN; w++) {
       uint32_t packed = dig[i * MSM_NWIN + w];
       uint32_t mag = packed & 0x1FFFFu;
       if (!mag) continue;
       uint32_t sign = (packed >> 17) & 1u;
-      uint32_t pos = atomicAdd(&off[w * MSM_NBUCK + (mag - 1)], 1u);
-      sorted[pos] = (uint32_t)i | (sign << 31);
+      uint32_t pos = atomicAdd(&off[bofs + w * MSM_NBUCK + (mag - 1)], 1u);
+      sorted[pos] = il | (sign << 31);
     }
   }
 }
@@ -237,9 +274,10 @@ __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
 
 // segment reduce: for window w, segment g over buckets [g*SEG, (g+1)*SEG):
 // partial = sum_{d in seg} (local_d+1)*B + (g*SEG)*W  where W = sum B.
-__global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* buckets, VestaJac* partials) {
+__global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* buckets, VestaJac* partials,
+                       u64 total_windows /* = MSM_NWIN * batch */) {
   u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x;
-  u64 ntot = (u64)MSM_NWIN * MSM_NSEG;
+  u64 ntot = total_windows * MSM_NSEG;
   for (; t < ntot; t += (u64)gridDim.x * blockDim.x) {
     u64 w = t / MSM_NSEG;
     u64 g = t % MSM_NSEG;
@@ -320,11 +358,14 @@ struct MsmWork {
   VestaJac* d_wsums = nullptr;  // MSM_NWIN window sums (combined on host)
   uint32_t* d_big = nullptr;    // big-bucket work list + count (phase 2)
   u64 cap_n = 0;
+  u64 cap_b = 1;
 };
 
-inline hipError_t msm_work_alloc(MsmWork& w, u64 n) {
-  if (w.cap_n >= n) return hipSuccess;
-  u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+inline hipError_t msm_work_alloc(MsmWork& w, u64 n_total, u64 batch = 1) {
+  if (w.cap_n >= n_total && w.cap_b >= batch) return hipSuccess;
+  if (n_total < w.cap_n) n_total = w.cap_n;
+  if (batch < w.cap_b) batch = w.cap_b;
+  u64 m = (u64)MSM_NWIN * MSM_NBUCK * batch;
   hipError_t e;
 #define TGW_FREE(p) \
   if (p) { hipFree(p); p = nullptr; }
@@ -332,19 +373,22 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n) {
   TGW_FREE(w.d_bsum) TGW_FREE(w.d_sorted) TGW_FREE(w.d_buckets) TGW_FREE(w.d_partials)
   TGW_FREE(w.d_wsums) TGW_FREE(w.d_big)
 #undef TGW_FREE
-  if ((e = hipMalloc(&w.d_dig, n * MSM_NWIN * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_dig, n_total * MSM_NWIN * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_hist, m * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_off, m * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_end, m * 4)) != hipSuccess) return e;
-  if ((e = hipMalloc(&w.d_bsum, ((m + 511) / 512) * 4)) != hipSuccess) return e;
-  if ((e = hipMalloc(&w.d_sorted, n * MSM_NWIN * 4)) != hipSuccess) return e;
+  u64 nb1 = (m + 511) / 512;
+  if ((e = hipMalloc(&w.d_bsum, (nb1 + (nb1 + 511) / 512 + 1) * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_sorted, n_total * MSM_NWIN * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_buckets, m * sizeof(VestaJac))) != hipSuccess) return e;
-  if ((e = hipMalloc(&w.d_partials, (u64)MSM_NWIN * MSM_NSEG * sizeof(VestaJac))) !=
-      hipSuccess)
+  if ((e = hipMalloc(&w.d_partials,
+                     (u64)MSM_NWIN * batch * MSM_NSEG * sizeof(VestaJac))) != hipSuccess)
     return e;
-  if ((e = hipMalloc(&w.d_wsums, MSM_NWIN * sizeof(VestaJac))) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_wsums, MSM_NWIN * batch * sizeof(VestaJac))) != hipSuccess)
+    return e;
   if ((e = hipMalloc(&w.d_big, (m + 1) * 4)) != hipSuccess) return e;
-  w.cap_n = n;
+  w.cap_n = n_total;
+  w.cap_b = batch;
   return hipSuccess;
 }
 

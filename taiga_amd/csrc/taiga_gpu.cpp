@@ -443,30 +443,23 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
   if ((e = msm_work_alloc(c->msm, n)) != hipSuccess) return set_err(c, "msm ws", e);
   {
     ProfScope total(c, P_MSM_TOTAL);
-    // (inner scopes only bracket sub-phases when profiling)
+    u64 m = (u64)MSM_NWIN * MSM_NBUCK;
     {
       ProfScope p(c, P_MSM_DIGITS);
-      u64 m = (u64)MSM_NWIN * MSM_NBUCK;
       hipMemsetAsync(c->msm.d_hist, 0, m * 4, c->stream);
       hipLaunchKernelGGL(k_digits, dim3(msm_grid(n)), dim3(256), 0, c->stream,
-                         c->d_scalars, n, c->msm.d_dig, c->msm.d_hist);
+                         c->d_scalars, n, n, c->msm.d_dig, c->msm.d_hist);
     }
-    u64 m = (u64)MSM_NWIN * MSM_NBUCK;
-    u64 nb = (m + 511) / 512;
     {
       ProfScope p(c, P_MSM_SCAN);
-      hipLaunchKernelGGL(k_scan_block, dim3((unsigned)nb), dim3(256), 0, c->stream,
-                         c->msm.d_hist, c->msm.d_off, c->msm.d_bsum, m);
-      hipLaunchKernelGGL(k_scan_sums, dim3(1), dim3(256), 0, c->stream, c->msm.d_bsum, nb);
-      hipLaunchKernelGGL(k_scan_add, dim3((unsigned)((m + 255) / 256)), dim3(256), 0,
-                         c->stream, c->msm.d_off, c->msm.d_bsum, m);
+      msm_scan(c->msm.d_hist, c->msm.d_off, c->msm.d_bsum, m, c->stream);
       hipMemcpyAsync(c->msm.d_hist, c->msm.d_off, m * 4, hipMemcpyDeviceToDevice,
                      c->stream);
     }
     {
       ProfScope p(c, P_MSM_SCATTER);
       hipLaunchKernelGGL(k_scatter, dim3(msm_grid(n)), dim3(256), 0, c->stream,
-                         c->msm.d_dig, n, c->msm.d_off, c->msm.d_sorted);
+                         c->msm.d_dig, n, n, c->msm.d_off, c->msm.d_sorted);
       hipMemcpyAsync(c->msm.d_end, c->msm.d_off, m * 4, hipMemcpyDeviceToDevice,
                      c->stream);
     }
@@ -492,7 +485,8 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     {
       ProfScope p(c, P_MSM_REDUCE);
       hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)MSM_NWIN * MSM_NSEG)),
-                         dim3(256), 0, c->stream, c->msm.d_buckets, c->msm.d_partials);
+                         dim3(256), 0, c->stream, c->msm.d_buckets, c->msm.d_partials,
+                         (u64)MSM_NWIN);
     }
     {
       ProfScope p(c, P_MSM_WSUM);
