@@ -258,12 +258,47 @@ inline void pprefix_prod(Fp* z, const Fp* r, long u, const Fp& init) {
   z[u] = acc;
 }
 
+// kate division q = a / (X - b), remainder dropped: q[i] = sum_{j>i} a[j] b^{j-i-1}.
+// Blocked two-pass form of the serial recurrence prev = a[i] + b*prev
+// (exact same values; ~2n muls, blocks parallel).
 inline void pkate_division(Fp* q, const Fp* a, long n, const Fp& b) {
-  Fp prev = a[n - 1];
-  for (long i = n - 2; i >= 0; i--) {
-    Fp ai = a[i];
-    q[i] = prev;
-    if (i >= 1) prev = fd_add(ai, fd_mul(b, prev));
+  const int T = 16;
+  long chunk = (n + T - 1) / T;
+  // pass 1: per block [lo,hi): local suffix Horner L_t = sum_{j in [lo,hi)} a[j] b^{j-lo}
+  Fp L[T], bp[T];  // bp = b^(hi-lo)
+  std::vector<Fp> asnap(a, a + n);  // allow q to alias a
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (int t = 0; t < T; t++) {
+    long lo = t * chunk, hi = std::min(lo + chunk, n);
+    Fp acc = fd_zero<FpCfg>();
+    for (long i = hi - 1; i >= lo; i--) acc = fd_add(fd_mul(acc, b), asnap[i]);
+    L[t] = acc;
+    bp[t] = fp_pow_small(b, hi > lo ? hi - lo : 0);
+  }
+  // serial: carry into each block from above: C_t = Horner of blocks > t
+  Fp C[T];
+  Fp acc = fd_zero<FpCfg>();
+  for (int t = T - 1; t >= 0; t--) {
+    C[t] = acc;
+    acc = fd_add(fd_mul(acc, bp[t]), L[t]);
+  }
+  // pass 2: per block, run the recurrence with the incoming carry
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (int t = 0; t < T; t++) {
+    long lo = t * chunk, hi = std::min(lo + chunk, n);
+    Fp prev = C[t];  // value of q at index hi-1's "incoming" state
+    for (long i = hi - 1; i >= lo; i--) {
+      if (i == n - 1) {
+        prev = asnap[i];
+        continue;  // q[n-1] does not exist; prev = a[n-1]
+      }
+      q[i] = prev;
+      prev = fd_add(asnap[i], fd_mul(b, prev));
+    }
   }
 }
 
