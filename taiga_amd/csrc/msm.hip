@@ -23,11 +23,20 @@
 
 namespace taiga {
 
-constexpr int MSM_C = 16;                    // window bits
-constexpr int MSM_NWIN = 16;                 // ceil(255/16)
-constexpr int MSM_NBUCK = 1 << (MSM_C - 1);  // 32768 buckets (digits 1..32768)
+// window size is a runtime parameter: c=16 for large MSMs (2^20 microbench),
+// c=12 for the prover's n=2^15 commits (11x smaller bucket space -> far
+// lower fixed cost per MSM). The result point is windowing-independent.
+constexpr int MSM_C_MAX = 16;
+constexpr int MSM_NWIN_MAX = 16;             // ceil(255/16)
+constexpr int MSM_NBUCK_MAX = 1 << (MSM_C_MAX - 1);
 constexpr int MSM_SEG = 16;                  // buckets per reduction segment
-constexpr int MSM_NSEG = MSM_NBUCK / MSM_SEG;
+struct MsmCfg {
+  int c, nwin, nbuck, nseg;
+};
+inline MsmCfg msm_cfg(long n) {
+  int c = n <= (1L << 16) ? 12 : 16;
+  return MsmCfg{c, (255 + c - 1) / c, 1 << (c - 1), (1 << (c - 1)) / MSM_SEG};
+}
 
 struct ScalarRepr {
   u64 l[4];
@@ -35,36 +44,37 @@ struct ScalarRepr {
 
 // digits: packed u32 = mag(17 bits) | sign<<17 ; mag==0 means skip.
 // hist[w * MSM_NBUCK + (mag-1)]++
-__global__ void __launch_bounds__(256) k_digits(const ScalarRepr* sc, u64 n, u64 nn, uint32_t* dig,
-                uint32_t* hist) {
+__global__ void __launch_bounds__(256) k_digits(const ScalarRepr* sc, u64 n, u64 nn, MsmCfg cfg,
+                uint32_t* dig, uint32_t* hist) {
   // n = total scalars (= nn * batch); histogram is per batch b = i / nn
-  const u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+  const u64 m = (u64)cfg.nwin * cfg.nbuck;
+  const uint32_t mask = (1u << cfg.c) - 1;
+  const uint32_t halfc = 1u << (cfg.c - 1);
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
     u64 bofs = (i / nn) * m;
     ScalarRepr s = sc[i];
     uint32_t carry = 0;
-#pragma unroll
-    for (int w = 0; w < MSM_NWIN; w++) {
-      int bit0 = w * MSM_C;
+    for (int w = 0; w < cfg.nwin; w++) {
+      int bit0 = w * cfg.c;
       int limb = bit0 >> 6, sh = bit0 & 63;
       u64 raw = s.l[limb] >> sh;
-      if (sh && limb < 3) raw |= s.l[limb + 1] << (64 - sh);
-      uint32_t d = (uint32_t)(raw & 0xFFFFu) + carry;
+      if (sh && limb < 3 && sh + cfg.c > 64) raw |= s.l[limb + 1] << (64 - sh);
+      uint32_t d = ((uint32_t)raw & mask) + carry;
       uint32_t sign = 0;
-      if (d > (1u << (MSM_C - 1))) {  // d in (2^15, 2^16]: take d - 2^16, carry
-        d = (1u << MSM_C) - d;
+      if (d > halfc) {  // take d - 2^c, carry into the next window
+        d = (1u << cfg.c) - d;
         sign = 1;
         carry = 1;
       } else {
         carry = 0;
       }
       uint32_t packed = d ? (d | (sign << 17)) : 0;
-      dig[i * MSM_NWIN + w] = packed;
-      if (d) atomicAdd(&hist[bofs + w * MSM_NBUCK + (d - 1)], 1u);
+      dig[i * cfg.nwin + w] = packed;
+      if (d) atomicAdd(&hist[bofs + (u64)w * cfg.nbuck + (d - 1)], 1u);
     }
-    // top window of a <2^255 scalar cannot carry out (bits 240..254 + carry
-    // <= 2^15 fits as a positive digit)
+    // the top window of a <2^255 scalar cannot carry out (its raw value +
+    // carry stays <= 2^(c-1) for c in {12,16})
   }
 }
 
@@ -154,20 +164,19 @@ inline void msm_scan(const uint32_t* in, uint32_t* out, uint32_t* bsum, u64 m,
 }
 
 // scatter: sorted[off[bucket]++] = i | sign<<31
-__global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, u64 nn, uint32_t* off,
-                 uint32_t* sorted) {
-  const u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+__global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, u64 nn, MsmCfg cfg,
+                 uint32_t* off, uint32_t* sorted) {
+  const u64 m = (u64)cfg.nwin * cfg.nbuck;
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
        i += (u64)gridDim.x * blockDim.x) {
     u64 bofs = (i / nn) * m;
     uint32_t il = (uint32_t)(i % nn);  // batch-local point index
-#pragma unroll
-    for (int w = 0; w < MSM_NWIN; w++) {
-      uint32_t packed = dig[i * MSM_NWIN + w];
+    for (int w = 0; w < cfg.nwin; w++) {
+      uint32_t packed = dig[i * cfg.nwin + w];
       uint32_t mag = packed & 0x1FFFFu;
       if (!mag) continue;
       uint32_t sign = (packed >> 17) & 1u;
-      uint32_t pos = atomicAdd(&off[bofs + w * MSM_NBUCK + (mag - 1)], 1u);
+      uint32_t pos = atomicAdd(&off[bofs + (u64)w * cfg.nbuck + (mag - 1)], 1u);
       sorted[pos] = il | (sign << 31);
     }
   }
@@ -275,13 +284,13 @@ __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
 // segment reduce: for window w, segment g over buckets [g*SEG, (g+1)*SEG):
 // partial = sum_{d in seg} (local_d+1)*B + (g*SEG)*W  where W = sum B.
 __global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* buckets, VestaJac* partials,
-                       u64 total_windows /* = MSM_NWIN * batch */) {
+                       u64 total_windows /* = nwin * batch */, MsmCfg cfg) {
   u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x;
-  u64 ntot = total_windows * MSM_NSEG;
+  u64 ntot = total_windows * cfg.nseg;
   for (; t < ntot; t += (u64)gridDim.x * blockDim.x) {
-    u64 w = t / MSM_NSEG;
-    u64 g = t % MSM_NSEG;
-    const VestaJac* B = buckets + w * MSM_NBUCK + g * MSM_SEG;
+    u64 w = t / cfg.nseg;
+    u64 g = t % cfg.nseg;
+    const VestaJac* B = buckets + w * (u64)cfg.nbuck + g * MSM_SEG;
     VestaJac run = jac_identity<FqCfg>();
     VestaJac tot = jac_identity<FqCfg>();
     for (int d = MSM_SEG - 1; d >= 0; d--) {
@@ -306,12 +315,12 @@ __global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* bucket
 // Jacobian window sum (the 16 window sums go to the host shim, which does
 // the O(1) 240-doubling Horner combine with the same TG_HD primitives —
 // that serial tail is host work, not a 1-lane GPU kernel).
-__global__ void __launch_bounds__(256) k_wsum(const VestaJac* partials, VestaJac* wsums) {
+__global__ void __launch_bounds__(256) k_wsum(const VestaJac* partials, VestaJac* wsums, MsmCfg cfg) {
   __shared__ VestaJac lds[256];
   int w = blockIdx.x;
   int t = threadIdx.x;  // 256 threads
   VestaJac acc = jac_identity<FqCfg>();
-  for (int g = t; g < MSM_NSEG; g += 256) acc = jac_add(acc, partials[w * MSM_NSEG + g]);
+  for (int g = t; g < cfg.nseg; g += 256) acc = jac_add(acc, partials[w * cfg.nseg + g]);
   lds[t] = acc;
   __syncthreads();
   for (int off = 128; off >= 1; off >>= 1) {
@@ -365,7 +374,7 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n_total, u64 batch = 1) {
   if (w.cap_n >= n_total && w.cap_b >= batch) return hipSuccess;
   if (n_total < w.cap_n) n_total = w.cap_n;
   if (batch < w.cap_b) batch = w.cap_b;
-  u64 m = (u64)MSM_NWIN * MSM_NBUCK * batch;
+  u64 m = (u64)MSM_NWIN_MAX * MSM_NBUCK_MAX * batch;
   hipError_t e;
 #define TGW_FREE(p) \
   if (p) { hipFree(p); p = nullptr; }
@@ -373,18 +382,19 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n_total, u64 batch = 1) {
   TGW_FREE(w.d_bsum) TGW_FREE(w.d_sorted) TGW_FREE(w.d_buckets) TGW_FREE(w.d_partials)
   TGW_FREE(w.d_wsums) TGW_FREE(w.d_big)
 #undef TGW_FREE
-  if ((e = hipMalloc(&w.d_dig, n_total * MSM_NWIN * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_dig, n_total * 22 * 4)) != hipSuccess) return e;  /* nwin<=22 (c=12) */
   if ((e = hipMalloc(&w.d_hist, m * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_off, m * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_end, m * 4)) != hipSuccess) return e;
   u64 nb1 = (m + 511) / 512;
   if ((e = hipMalloc(&w.d_bsum, (nb1 + (nb1 + 511) / 512 + 1) * 4)) != hipSuccess) return e;
-  if ((e = hipMalloc(&w.d_sorted, n_total * MSM_NWIN * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_sorted, n_total * 22 * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_buckets, m * sizeof(VestaJac))) != hipSuccess) return e;
-  if ((e = hipMalloc(&w.d_partials,
-                     (u64)MSM_NWIN * batch * MSM_NSEG * sizeof(VestaJac))) != hipSuccess)
+  if ((e = hipMalloc(&w.d_partials, (u64)MSM_NWIN_MAX * 2 * batch * MSM_NBUCK_MAX /
+                                        MSM_SEG * sizeof(VestaJac))) != hipSuccess)
     return e;
-  if ((e = hipMalloc(&w.d_wsums, MSM_NWIN * batch * sizeof(VestaJac))) != hipSuccess)
+  if ((e = hipMalloc(&w.d_wsums, MSM_NWIN_MAX * 2 * batch * sizeof(VestaJac))) !=
+      hipSuccess)
     return e;
   if ((e = hipMalloc(&w.d_big, (m + 1) * 4)) != hipSuccess) return e;
   w.cap_n = n_total;
@@ -401,11 +411,11 @@ static inline int msm_grid(u64 work, int block = 256) {
 
 // host-side final combine: acc = sum_w 2^(16w) * wsum[w]  (Horner, ~240
 // doublings of O(1) work — the same TG_HD primitives as the kernels)
-inline VestaAff msm_host_combine(const VestaJac wsums[MSM_NWIN]) {
+inline VestaAff msm_host_combine(const VestaJac* wsums, const MsmCfg& cfg) {
   VestaJac acc = jac_identity<FqCfg>();
-  for (int w = MSM_NWIN - 1; w >= 0; w--) {
-    if (w != MSM_NWIN - 1)
-      for (int b = 0; b < MSM_C; b++) acc = jac_dbl(acc);
+  for (int w = cfg.nwin - 1; w >= 0; w--) {
+    if (w != cfg.nwin - 1)
+      for (int b = 0; b < cfg.c; b++) acc = jac_dbl(acc);
     acc = jac_add(acc, wsums[w]);
   }
   return jac_to_aff(acc);

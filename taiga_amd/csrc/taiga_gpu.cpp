@@ -441,14 +441,15 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
   }
   hipError_t e;
   if ((e = msm_work_alloc(c->msm, n)) != hipSuccess) return set_err(c, "msm ws", e);
+  MsmCfg cfg = msm_cfg((long)n);
   {
     ProfScope total(c, P_MSM_TOTAL);
-    u64 m = (u64)MSM_NWIN * MSM_NBUCK;
+    u64 m = (u64)cfg.nwin * cfg.nbuck;
     {
       ProfScope p(c, P_MSM_DIGITS);
       hipMemsetAsync(c->msm.d_hist, 0, m * 4, c->stream);
       hipLaunchKernelGGL(k_digits, dim3(msm_grid(n)), dim3(256), 0, c->stream,
-                         c->d_scalars, n, n, c->msm.d_dig, c->msm.d_hist);
+                         c->d_scalars, n, n, cfg, c->msm.d_dig, c->msm.d_hist);
     }
     {
       ProfScope p(c, P_MSM_SCAN);
@@ -459,7 +460,7 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     {
       ProfScope p(c, P_MSM_SCATTER);
       hipLaunchKernelGGL(k_scatter, dim3(msm_grid(n)), dim3(256), 0, c->stream,
-                         c->msm.d_dig, n, n, c->msm.d_off, c->msm.d_sorted);
+                         c->msm.d_dig, n, n, cfg, c->msm.d_off, c->msm.d_sorted);
       hipMemcpyAsync(c->msm.d_end, c->msm.d_off, m * 4, hipMemcpyDeviceToDevice,
                      c->stream);
     }
@@ -484,21 +485,22 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     }
     {
       ProfScope p(c, P_MSM_REDUCE);
-      hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)MSM_NWIN * MSM_NSEG)),
+      hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)cfg.nwin * cfg.nseg)),
                          dim3(256), 0, c->stream, c->msm.d_buckets, c->msm.d_partials,
-                         (u64)MSM_NWIN);
+                         (u64)cfg.nwin, cfg);
     }
     {
       ProfScope p(c, P_MSM_WSUM);
-      hipLaunchKernelGGL(k_wsum, dim3(MSM_NWIN), dim3(256), 0, c->stream,
-                         c->msm.d_partials, c->msm.d_wsums);
+      hipLaunchKernelGGL(k_wsum, dim3(cfg.nwin), dim3(256), 0, c->stream,
+                         c->msm.d_partials, c->msm.d_wsums, cfg);
     }
   }
-  VestaJac wsums[MSM_NWIN];
-  hipMemcpyAsync(wsums, c->msm.d_wsums, sizeof(wsums), hipMemcpyDeviceToHost, c->stream);
+  VestaJac wsums[22];
+  hipMemcpyAsync(wsums, c->msm.d_wsums, sizeof(VestaJac) * cfg.nwin,
+                 hipMemcpyDeviceToHost, c->stream);
   hipError_t es = hipStreamSynchronize(c->stream);
   if (es != hipSuccess) return set_err(c, "msm run", es);
-  VestaAff r = msm_host_combine(wsums);
+  VestaAff r = msm_host_combine(wsums, cfg);
   if (aff_is_identity(r)) {
     memset(out_xy, 0, 64);
   } else {
