@@ -126,6 +126,57 @@ __global__ void __launch_bounds__(256) k_ntt_fused(Fd<C>* a, const Fd<C>* tw, in
   }
 }
 
+// 2-D fused LDS stages for strided passes: each block processes RB
+// consecutive residues r, so global accesses are RB*32-byte contiguous
+// chunks (the 1-residue variant gathers single 32-B elements at stride
+// 2^s0 — ~4.5x slower per pass at k=22). F local stages; span = 2^s0 must
+// be a multiple of RB. LDS rows padded (+1 element) to spread bank groups.
+template <class C, int F, int RB>
+__global__ void __launch_bounds__(256) k_ntt_fused2(Fd<C>* a, const Fd<C>* tw, int k,
+                                                    int s0) {
+  constexpr int TILE = 1 << F;
+  __shared__ Fd<C> lds[(TILE + 1) * RB];
+  u64 n = 1ULL << k;
+  u64 span = 1ULL << s0;
+  u64 groups_per_span = span / RB;
+  u64 seg_len = span << F;
+  u64 nseg = n / seg_len;
+  u64 ntiles = nseg * groups_per_span;
+  for (u64 tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    u64 seg = tile / groups_per_span;
+    u64 r0 = (tile % groups_per_span) * RB;
+    u64 base = seg * seg_len + r0;
+    for (int t = threadIdx.x; t < TILE * RB; t += blockDim.x) {
+      int m = t / RB, rr = t % RB;
+      lds[rr * (TILE + 1) + m] = a[base + (u64)m * span + rr];
+    }
+    __syncthreads();
+    for (int ls = 1; ls <= F; ls++) {
+      int s = s0 + ls;
+      u64 half = 1ULL << (ls - 1);
+      int tshift = k - s;
+      for (int t = threadIdx.x; t < (TILE / 2) * RB; t += blockDim.x) {
+        int j = t / RB, rr = t % RB;
+        u64 grp = (u64)j >> (ls - 1);
+        u64 kk = (u64)j & (half - 1);
+        int i0 = (int)((grp << ls) | kk);
+        int i1 = i0 + (int)half;
+        u64 kkg = (kk << s0) | (r0 + rr);
+        Fd<C> tv = fd_mul(lds[rr * (TILE + 1) + i1], tw[kkg << tshift]);
+        Fd<C> lo = lds[rr * (TILE + 1) + i0];
+        lds[rr * (TILE + 1) + i1] = fd_sub(lo, tv);
+        lds[rr * (TILE + 1) + i0] = fd_add(lo, tv);
+      }
+      __syncthreads();
+    }
+    for (int t = threadIdx.x; t < TILE * RB; t += blockDim.x) {
+      int m = t / RB, rr = t % RB;
+      a[base + (u64)m * span + rr] = lds[rr * (TILE + 1) + m];
+    }
+    __syncthreads();
+  }
+}
+
 // pointwise scale (by a constant) and/or from-mont conversion
 template <class C>
 __global__ void __launch_bounds__(256) k_scale(Fd<C>* a, u64 n, Fd<C> c, int do_scale, int from_mont_flag) {
@@ -195,17 +246,37 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
     hipLaunchKernelGGL(k_bitrev_load<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream,
                        d_tmp, d_a, k, 0, nullptr);
   }
-  constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB)
+  constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB), first pass only
+  constexpr int RB = 4;    // residues per block in the strided passes
   int s = 1;
   while (s <= k) {
     int remaining = k - s + 1;
-    if (remaining >= FUSE && k >= FUSE) {
+    if (s == 1 && remaining >= FUSE) {
+      // first pass: span 1, tiles are contiguous -> 1-D variant
       auto sc = prof(1);
       u64 ntiles = n >> FUSE;
       hipLaunchKernelGGL((k_ntt_fused<FpCfg, FUSE>),
                          dim3(ntiles > 2048 ? 2048 : (unsigned)ntiles), dim3(256), 0,
                          stream, d_tmp, tw, k, s - 1);
       s += FUSE;
+      continue;
+    }
+    int f = remaining < 8 ? remaining : 8;
+    u64 span = 1ULL << (s - 1);
+    if (f >= 2 && span >= RB) {
+      auto sc = prof(1);
+      u64 ntiles = n >> f;
+      unsigned grid = ntiles > 2048 ? 2048 : (unsigned)ntiles;
+      switch (f) {
+        case 8: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 8, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+        case 7: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 7, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+        case 6: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 6, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+        case 5: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 5, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+        case 4: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 4, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+        case 3: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 3, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+        default: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 2, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
+      }
+      s += f;
       continue;
     }
     {
