@@ -68,7 +68,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--workload", choices=["proof", "msm", "ntt"], default="proof")
-    ap.add_argument("--streams", type=int, default=2,
+    ap.add_argument("--streams", type=int, default=6,
                     help="concurrent proving contexts per GPU (proof workload)")
     args = ap.parse_args()
 
