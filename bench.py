@@ -37,7 +37,10 @@ def _cpu_quota():
     return os.cpu_count() or 8
 
 
-N_CORES = min(_cpu_quota(), os.cpu_count() or 8)
+# divide the quota among co-located ranks (torchrun sets LOCAL_WORLD_SIZE);
+# 8 ranks x quota-sized spinning teams would re-create the throttle stalls
+_local_world = int(os.environ.get("LOCAL_WORLD_SIZE", os.environ.get("WORLD_SIZE", "1")))
+N_CORES = max(1, min(_cpu_quota(), os.cpu_count() or 8) // max(1, _local_world))
 os.environ.setdefault("OMP_NUM_THREADS", str(N_CORES))
 os.environ.setdefault("OMP_WAIT_POLICY", "PASSIVE")
 
@@ -68,8 +71,9 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--workload", choices=["proof", "msm", "ntt"], default="proof")
-    ap.add_argument("--streams", type=int, default=6,
-                    help="concurrent proving contexts per GPU (proof workload)")
+    ap.add_argument("--streams", type=int, default=0,
+                    help="concurrent proving contexts per GPU (proof workload); "
+                         "0 = auto (6 single-rank, 2 when ranks share the host quota)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -106,7 +110,8 @@ def main():
         golden = pathlib.Path(REPO) / "tests" / "golden"
         srs_bytes = (golden / "params_15").read_bytes()
         desc_bytes = (golden / "cs1.desc").read_bytes()
-        C = max(1, args.streams)
+        C = args.streams if args.streams > 0 else (6 if _local_world <= 1 else 2)
+        args.streams = C
         ctxs = [gpu] + [taiga_amd.TaigaGpu(local_rank) for _ in range(C - 1)]
         for g in ctxs:
             g.load_srs(srs_bytes)
