@@ -108,6 +108,13 @@ int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len);
 int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], const uint8_t rng_seed[32],
                     uint8_t* proof_out, size_t cap, size_t* out_len);
+/* raw-witness variant: instance = n_instance_rows x 32B canonical reprs,
+ * advice = n_advice x 2^k x 32B canonical column-major (rows beyond
+ * usable are replaced by prover blinding). Same proof bytes as the seeded
+ * path when fed the generated witness. */
+int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* advice,
+                        const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
+                        size_t* out_len);
 /* verify one proof (replaces Proof::verify / plonk::verify_proof with the
  * SingleVerifier strategy — proof.rs:45-54). Returns TG_OK iff valid. */
 int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* proof,

@@ -2754,3 +2754,29 @@ long orc_dbg_perm_base(const uint8_t inst_seed[32], const uint8_t wit_seed[32],
     g_hmask = save;
     return bad;
 }
+
+/* export the CS1-generated witness as canonical bytes (tests / raw-path
+ * parity): instance n_instance_rows x 32B, advice n_advice x n x 32B. */
+int orc_cs1_export_witness(const uint8_t inst_seed[32], const uint8_t wit_seed[32],
+                           uint8_t* inst_out, uint8_t* advice_out) {
+    if (!g_pk) return -1;
+    Desc* d = g_pk->d;
+    long n = d->n;
+    fd_limbs* inst = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
+    orc_cs1_instance(d, inst_seed, inst);
+    for (int r = 0; r < d->n_instance_rows; r++) fd_to_bytes(inst_out + 32 * r, inst[r], FP);
+    fd_limbs* adv[16];
+    for (int c2 = 0; c2 < d->n_advice; c2++)
+        adv[c2] = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
+    orc_cs1_witness(d, wit_seed, inst, adv);
+    for (int c2 = 0; c2 < d->n_advice; c2++) {
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+        for (long i = 0; i < n; i++)
+            fd_to_bytes(advice_out + ((size_t)c2 * n + i) * 32, adv[c2][i], FP);
+        free(adv[c2]);
+    }
+    free(inst);
+    return 0;
+}

@@ -72,6 +72,10 @@ def load_library():
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
             ctypes.c_char_p, ctypes.c_size_t, ctypes.POINTER(ctypes.c_size_t),
         ]
+        lib.tg_create_proof_raw.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
+            ctypes.c_char_p, ctypes.c_size_t, ctypes.POINTER(ctypes.c_size_t),
+        ]
         lib.tg_verify_proof.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
         ]
@@ -179,6 +183,13 @@ class TaigaGpu:
         out_len = ctypes.c_size_t()
         self._ck(self._lib.tg_create_proof(
             self._h, inst_seed, wit_seed, rng_seed, out, len(out), ctypes.byref(out_len)))
+        return out.raw[: out_len.value]
+
+    def create_proof_raw(self, instance: bytes, advice: bytes, rng_seed: bytes) -> bytes:
+        out = ctypes.create_string_buffer(1 << 16)
+        out_len = ctypes.c_size_t()
+        self._ck(self._lib.tg_create_proof_raw(
+            self._h, instance, advice, rng_seed, out, len(out), ctypes.byref(out_len)))
         return out.raw[: out_len.value]
 
     def verify_proof(self, inst_seed: bytes, proof: bytes) -> bool:

@@ -673,3 +673,25 @@ int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* pro
 }
 
 }  /* extern "C" */
+
+extern "C" {
+
+/* raw-witness proving (SURVEY §8b witness-blob shape): instance =
+ * n_instance_rows x 32B canonical reprs; advice = n_advice x 2^k x 32B
+ * canonical, column-major (rows beyond usable = n-(bf+1) are replaced by
+ * blinding). */
+int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* advice,
+                        const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
+                        size_t* out_len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  std::vector<uint8_t> proof;
+  int rc = pprove_raw(c, *c->ppk, instance, advice, rng_seed, proof);
+  if (rc != 0) return rc;
+  if (proof.size() > cap) return TG_ERR_BADARG;
+  memcpy(proof_out, proof.data(), proof.size());
+  *out_len = proof.size();
+  return TG_OK;
+}
+
+}  /* extern "C" */

@@ -82,3 +82,18 @@ def test_gpu_proof_verifies_and_seeds_differ(oracle_pk, gpu_pk):
         assert oracle_pk.orc_verify_cs1(inst, p, len(p)) == 0
         # and rejects it against a different instance
         assert oracle_pk.orc_verify_cs1(bytes([99]) + bytes(31), p, len(p)) != 0
+
+
+def test_raw_witness_path_matches_seeded(oracle_pk, gpu_pk):
+    """tg_create_proof_raw fed the exported CS1 witness must produce the
+    SAME proof bytes as the seeded path (proof depends only on
+    SRS/desc/witness/rng — the §8b witness-blob ABI shape)."""
+    n = 1 << 15
+    inst = ctypes.create_string_buffer(9 * 32)
+    adv = ctypes.create_string_buffer(10 * n * 32)
+    assert oracle_pk.orc_cs1_export_witness(INST, WIT, inst, adv) == 0
+    raw_proof = gpu_pk.create_proof_raw(inst.raw, adv.raw, RNG)
+    seeded_proof = gpu_pk.create_proof(INST, WIT, RNG)
+    assert raw_proof == seeded_proof
+    # and still verifies
+    assert gpu_pk.verify_proof(INST, raw_proof)
