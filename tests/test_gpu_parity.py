@@ -164,3 +164,35 @@ def test_msm_commit_lagrange_shape(gpu, params15):
         got = gpu.msm(bytes(sc), base_set=2)
         exp = oc.decompress(oc.FQ, params15[4 + 32 * (n + i) : 4 + 32 * (n + i) + 32])
         assert got == exp
+
+
+def test_ntt_parity_k18(gpu):
+    # extended-domain size used by the prover (2^18)
+    rng = random.Random(1818)
+    k = 18
+    data = rand_scalars(rng, 1 << k)
+    assert gpu.ntt(data, k) == oc.ntt(oc.FP, 0, k, data)
+
+
+def test_msm_parity_2e17(gpu):
+    # mid-size MSM parity vs oracle (between prover 2^15 and bench 2^20)
+    rng = random.Random(1717)
+    n = 1 << 17
+    pts = oc.gen_bases(n, 999)
+    sc = rand_scalars(rng, n)
+    gpu.bases_upload(pts)
+    assert gpu.msm(sc, base_set=0) == oc.msm(oc.FQ, sc, pts)
+
+
+def test_msm_skewed_scalars(gpu, params15):
+    """duplicate-heavy scalar vectors (the grand-product tail shape) hit the
+    wave-per-bucket phase-2 kernel; parity must hold there too."""
+    gpu.load_srs(params15)
+    n = 1 << 15
+    rng = random.Random(555)
+    v = rng.randrange(pp.P)
+    sc = bytearray()
+    for i in range(n):
+        sc += (v if i > 100 else rng.randrange(pp.P)).to_bytes(32, "little")
+    g = oc.decompress(oc.FQ, params15[4 : 4 + 32 * n])
+    assert gpu.msm(bytes(sc), base_set=1) == oc.msm(oc.FQ, bytes(sc), g)

@@ -97,3 +97,11 @@ def test_raw_witness_path_matches_seeded(oracle_pk, gpu_pk):
     assert raw_proof == seeded_proof
     # and still verifies
     assert gpu_pk.verify_proof(INST, raw_proof)
+
+
+def test_proof_parity_more_seeds(oracle_pk, gpu_pk):
+    for s in (101, 202):
+        inst = bytes([s % 251]) + bytes(31)
+        wit = bytes([(s * 3) % 251]) + bytes(31)
+        rng = bytes([(s * 7) % 251]) + bytes(31)
+        assert gpu_pk.create_proof(inst, wit, rng) == oracle_prove(oracle_pk, inst, wit, rng)
