@@ -233,79 +233,6 @@ __global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, co
   }
 }
 
-// 2-buckets-per-thread fast accumulation: two independent add chains per
-// lane give the scheduler ILP to cover the ~1.5k-cycle dependent latency of
-// a mixed add (single-chain version is issue/latency-bound at ~4 waves/SIMD)
-__global__ void __launch_bounds__(256, 1) k_bucket_acc_fast2(const uint32_t* start,
-                              const uint32_t* end, const uint32_t* sorted,
-                              const VestaAff* pts, VestaJac* buckets,
-                              u64 nbuckets_total, uint32_t* big_list,
-                              uint32_t* big_count) {
-  u64 half = (nbuckets_total + 1) / 2;
-  for (u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x; t < half;
-       t += (u64)gridDim.x * blockDim.x) {
-    u64 b0 = t, b1 = t + half;
-    uint32_t s0 = start[b0], e0 = end[b0];
-    uint32_t s1 = 0, e1 = 0;
-    if (b1 < nbuckets_total) {
-      s1 = start[b1];
-      e1 = end[b1];
-    }
-    bool big0 = e0 - s0 > MSM_BIG_BUCKET, big1 = b1 < nbuckets_total && e1 - s1 > MSM_BIG_BUCKET;
-    if (big0) {
-      uint32_t slot = atomicAdd(big_count, 1u);
-      big_list[slot] = (uint32_t)b0;
-      s0 = e0 = 0;
-    }
-    if (big1) {
-      uint32_t slot = atomicAdd(big_count, 1u);
-      big_list[slot] = (uint32_t)b1;
-      s1 = e1 = 0;
-    }
-    VestaJac a0, a1;
-    bool h0 = s0 < e0, h1 = s1 < e1;
-    if (h0) {
-      uint32_t ent = sorted[s0++];
-      VestaAff p = pts[ent & 0x7FFFFFFFu];
-      if (ent >> 31) p = aff_neg_fast(p);
-      a0.x = p.x; a0.y = p.y; a0.z = fd_one_mont<FqCfg>();
-    } else {
-      a0 = jac_identity<FqCfg>();
-    }
-    if (h1) {
-      uint32_t ent = sorted[s1++];
-      VestaAff p = pts[ent & 0x7FFFFFFFu];
-      if (ent >> 31) p = aff_neg_fast(p);
-      a1.x = p.x; a1.y = p.y; a1.z = fd_one_mont<FqCfg>();
-    } else {
-      a1 = jac_identity<FqCfg>();
-    }
-    while (s0 < e0 && s1 < e1) {
-      uint32_t ea = sorted[s0++], eb = sorted[s1++];
-      VestaAff pa = pts[ea & 0x7FFFFFFFu];
-      VestaAff pb = pts[eb & 0x7FFFFFFFu];
-      if (ea >> 31) pa = aff_neg_fast(pa);
-      if (eb >> 31) pb = aff_neg_fast(pb);
-      jac_add_aff_fast(a0, pa);
-      jac_add_aff_fast(a1, pb);
-    }
-    for (; s0 < e0; s0++) {
-      uint32_t ent = sorted[s0];
-      VestaAff p = pts[ent & 0x7FFFFFFFu];
-      if (ent >> 31) p = aff_neg_fast(p);
-      jac_add_aff_fast(a0, p);
-    }
-    for (; s1 < e1; s1++) {
-      uint32_t ent = sorted[s1];
-      VestaAff p = pts[ent & 0x7FFFFFFFu];
-      if (ent >> 31) p = aff_neg_fast(p);
-      jac_add_aff_fast(a1, p);
-    }
-    if (!big0) buckets[b0] = a0;
-    if (b1 < nbuckets_total && !big1) buckets[b1] = a1;
-  }
-}
-
 // phase 2: one 64-lane wave per big bucket; lane-strided partials + LDS tree
 template <bool SAFE>
 __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
