@@ -247,7 +247,7 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
                        d_tmp, d_a, k, 0, nullptr);
   }
   constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB), first pass only
-  constexpr int RB = 4;    // residues per block in the strided passes
+  constexpr int RB = 8;    // residues per block in the strided passes (256-B gathers)
   int s = 1;
   while (s <= k) {
     int remaining = k - s + 1;
@@ -261,14 +261,13 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
       s += FUSE;
       continue;
     }
-    int f = remaining < 8 ? remaining : 8;
+    int f = remaining < 7 ? remaining : 7;  // RB=8 at F=7: 33 KiB LDS
     u64 span = 1ULL << (s - 1);
     if (f >= 2 && span >= RB) {
       auto sc = prof(1);
       u64 ntiles = n >> f;
       unsigned grid = ntiles > 2048 ? 2048 : (unsigned)ntiles;
       switch (f) {
-        case 8: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 8, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
         case 7: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 7, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
         case 6: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 6, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
         case 5: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 5, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
