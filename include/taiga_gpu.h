@@ -104,7 +104,11 @@ int tg_poly_download(tg_ctx* ctx, uint8_t* poly, uint32_t k);
  * (ChaCha20 — BASELINE.md convention; the reference draws from caller RNG,
  * proof.rs:30). Proof bytes are bit-identical to the CPU oracle on the same
  * seeds (tests/test_prover_parity.py). */
+/* returns the slot id (>= 0) of the cached PK and makes it active; PKs are
+ * cached per ctx (SURVEY §8f-1 — the reference's generic macro re-keygens
+ * every proof, resource_logic_circuit.rs:578-580). */
 int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len);
+int tg_select_key(tg_ctx* ctx, int slot);
 int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], const uint8_t rng_seed[32],
                     uint8_t* proof_out, size_t cap, size_t* out_len);

@@ -68,6 +68,7 @@ def load_library():
         ]
         lib.tg_poly_download.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint32]
         lib.tg_keygen.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
+        lib.tg_select_key.argtypes = [ctypes.c_void_p, ctypes.c_int]
         lib.tg_create_proof.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
             ctypes.c_char_p, ctypes.c_size_t, ctypes.POINTER(ctypes.c_size_t),
@@ -175,8 +176,14 @@ class TaigaGpu:
         return buf.raw
 
     # --- proving ---
-    def keygen(self, desc: bytes):
-        self._ck(self._lib.tg_keygen(self._h, desc, len(desc)))
+    def keygen(self, desc: bytes) -> int:
+        rc = self._lib.tg_keygen(self._h, desc, len(desc))
+        if rc < 0:
+            raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+        return rc  # slot id
+
+    def select_key(self, slot: int):
+        self._ck(self._lib.tg_select_key(self._h, slot))
 
     def create_proof(self, inst_seed: bytes, wit_seed: bytes, rng_seed: bytes) -> bytes:
         out = ctypes.create_string_buffer(1 << 16)

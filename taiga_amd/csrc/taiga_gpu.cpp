@@ -166,7 +166,9 @@ struct Ctx {
 
   MsmWork msm;
   NttPlan ntt;
-  struct PPk* ppk = nullptr;  // proving key (prover_gpu.inc)
+  struct PPk* ppk = nullptr;            // active proving key (prover_gpu.inc)
+  std::vector<struct PPk*> ppk_slots;   // PK cache: one entry
+                                        // per tg_keygen call; slot id = index
 
   bool prof = false;
   ProfCounter prof_c[PROF_N];
@@ -247,11 +249,12 @@ void tg_destroy(tg_ctx* ctx) {
   Ctx* c = (Ctx*)ctx;
   if (!c) return;
   hipStreamSynchronize(c->stream);
-  if (c->ppk) {
-    pdev_free(c->ppk->pd);
-    delete c->ppk;
-    c->ppk = nullptr;
+  for (PPk* pk : c->ppk_slots) {
+    pdev_free(pk->pd);
+    delete pk;
   }
+  c->ppk_slots.clear();
+  c->ppk = nullptr;
 #define TGF(p) \
   if (p) hipFree(p)
   TGF(c->d_g); TGF(c->d_gl); TGF(c->d_bases); TGF(c->d_scalars);
@@ -609,17 +612,30 @@ int tg_ntt_fp(tg_ctx* ctx, int dir, uint32_t k, int coset, uint8_t* poly) {
 /* ---- prover ABI (implementation in prover_gpu.inc) ---- */
 extern "C" {
 
+/* builds a proving key and caches it; returns the non-negative slot id
+ * (also leaves it active for the slotless entry points). PKs are cached for
+ * the ctx lifetime — the reference's generic-macro path re-keygens per
+ * proof (resource_logic_circuit.rs:578-580); the cache is §8f-1's fix. */
 int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len) {
   Ctx* c = (Ctx*)ctx;
   if (c->k < 0) return TG_ERR_NOSRS;
-  if (c->ppk) { delete c->ppk; c->ppk = nullptr; }
-  c->ppk = new PPk();
-  int rc = ppk_keygen(c, *c->ppk, desc, desc_len);
+  PPk* pk = new PPk();
+  int rc = ppk_keygen(c, *pk, desc, desc_len);
   if (rc != 0) {
-    delete c->ppk;
-    c->ppk = nullptr;
+    delete pk;
+    return rc;
   }
-  return rc;
+  c->ppk_slots.push_back(pk);
+  c->ppk = pk;
+  return (int)c->ppk_slots.size() - 1;
+}
+
+/* select a previously built proving key by slot id */
+int tg_select_key(tg_ctx* ctx, int slot) {
+  Ctx* c = (Ctx*)ctx;
+  if (slot < 0 || (size_t)slot >= c->ppk_slots.size()) return TG_ERR_BADARG;
+  c->ppk = c->ppk_slots[slot];
+  return TG_OK;
 }
 
 int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
