@@ -105,3 +105,21 @@ def test_proof_parity_more_seeds(oracle_pk, gpu_pk):
         wit = bytes([(s * 3) % 251]) + bytes(31)
         rng = bytes([(s * 7) % 251]) + bytes(31)
         assert gpu_pk.create_proof(inst, wit, rng) == oracle_prove(oracle_pk, inst, wit, rng)
+
+def test_key_slots(gpu_pk):
+    """PK cache: a second keygen gets a new slot, both slots
+    stay selectable, and proofs from either slot of the same circuit are
+    identical. Invalid slots are rejected."""
+    import taiga_amd
+    desc = open(os.path.join(GOLDEN, "cs1.desc"), "rb").read()
+    slot1 = gpu_pk.keygen(desc)
+    assert slot1 == 1  # module fixture already built slot 0
+    p1 = gpu_pk.create_proof(INST, WIT, RNG)
+    gpu_pk.select_key(0)
+    p0 = gpu_pk.create_proof(INST, WIT, RNG)
+    assert p0 == p1
+    with pytest.raises(taiga_amd.TaigaGpuError):
+        gpu_pk.select_key(2)
+    with pytest.raises(taiga_amd.TaigaGpuError):
+        gpu_pk.select_key(-1)
+    gpu_pk.select_key(slot1)
