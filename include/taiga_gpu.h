@@ -123,6 +123,15 @@ int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* adv
  * SingleVerifier strategy — proof.rs:45-54). Returns TG_OK iff valid. */
 int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* proof,
                     size_t proof_len);
+/* batch verification (halo2's BatchVerifier accumulation for
+ * a bundle's proofs — shielded_ptx.rs:137-153): all m final IPA checks
+ * collapse into one randomly-weighted combined check sharing a single
+ * g-sized GPU MSM. inst_seeds = m x 32B; proofs = concatenated proof bytes
+ * with per-proof lengths in proof_lens. All proofs must use the active key.
+ * TG_OK iff all valid; -1 if the combined check fails; -1xx on the first
+ * malformed proof. */
+int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
+                    const uint8_t* proofs, const size_t* proof_lens);
 /* blake2b-256 of the generated advice matrix (witness-spec cross-check) */
 int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], uint8_t out[32]);

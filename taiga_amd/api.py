@@ -80,6 +80,10 @@ def load_library():
         lib.tg_verify_proof.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
         ]
+        lib.tg_verify_batch.argtypes = [
+            ctypes.c_void_p, ctypes.c_size_t, ctypes.c_char_p, ctypes.c_char_p,
+            ctypes.POINTER(ctypes.c_size_t),
+        ]
         lib.tg_witness_hash.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
         ]
@@ -204,6 +208,20 @@ class TaigaGpu:
         if rc == 0:
             return True
         # -1 = final check failed; -1xx = malformed/truncated transcript
+        if rc == -1 or rc <= -100:
+            return False
+        raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+
+    def verify_batch(self, items) -> bool:
+        """Batch-verify [(inst_seed, proof), ...] in one combined IPA check
+        (one shared g-sized GPU MSM for all proofs — SURVEY §8f-3)."""
+        m = len(items)
+        seeds = b"".join(i for i, _ in items)
+        proofs = b"".join(p for _, p in items)
+        lens = (ctypes.c_size_t * m)(*[len(p) for _, p in items])
+        rc = self._lib.tg_verify_batch(self._h, m, seeds, proofs, lens)
+        if rc == 0:
+            return True
         if rc == -1 or rc <= -100:
             return False
         raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())

@@ -688,6 +688,44 @@ int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* pro
   return pverify(c, *c->ppk, inst_seed, proof, proof_len);
 }
 
+/* batch verification (the halo2 BatchVerifier / Guard
+ * accumulation pattern behind proof.rs:45-54): each proof contributes a
+ * final-check guard; guards are combined with random weights so that all
+ * m IPA final checks collapse into ONE g-sized GPU MSM plus a few hundred
+ * variable points. Weights are derived by hashing the whole batch (a
+ * prover cannot predict them). All proofs must be for the active key.
+ * Returns TG_OK iff every proof is valid; -1 if the combined check fails;
+ * -1xx on the first structurally malformed proof. */
+int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
+                    const uint8_t* proofs, const size_t* proof_lens) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  if (m == 0 || m > 4096 || !inst_seeds || !proofs || !proof_lens) return TG_ERR_BADARG;
+  std::vector<PVGuard> gds(m);
+  size_t off = 0;
+  for (size_t i = 0; i < m; i++) {
+    int rc = pverify_guard(c, *c->ppk, inst_seeds + 32 * i, proofs + off,
+                           proof_lens[i], gds[i]);
+    if (rc != 0) return rc;
+    off += proof_lens[i];
+  }
+  /* weights: rho_0 = 1, rho_i = DRBG(Blake2b(seeds ‖ proofs)) field draws */
+  std::vector<Fp> rho(m);
+  rho[0] = fd_one_mont<FpCfg>();
+  if (m > 1) {
+    Blake2b h(32, (const uint8_t*)"TaigaGPU-BatchVf");
+    uint64_t mle = (uint64_t)m;
+    h.update((const uint8_t*)&mle, 8);
+    h.update(inst_seeds, 32 * m);
+    h.update(proofs, off);
+    uint8_t seed[32];
+    h.final(seed);
+    Drbg rng(seed);
+    for (size_t i = 1; i < m; i++) rho[i] = rng.field<FpCfg>();
+  }
+  return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
+}
+
 }  /* extern "C" */
 
 extern "C" {

@@ -123,3 +123,41 @@ def test_key_slots(gpu_pk):
     with pytest.raises(taiga_amd.TaigaGpuError):
         gpu_pk.select_key(-1)
     gpu_pk.select_key(slot1)
+
+def test_batch_verify(oracle_pk, gpu_pk):
+    """tg_verify_batch: one combined IPA check accepts a
+    bundle of valid proofs (GPU- and oracle-produced, distinct instances)
+    and rejects the bundle when any single proof is tampered with or
+    mismatched against its instance."""
+    gpu_pk.select_key(0)
+    items = []
+    for s in range(4):
+        inst = bytes([40 + s]) + bytes(31)
+        wit = bytes([50 + s]) + bytes(31)
+        rng = bytes([60 + s]) + bytes(31)
+        p = gpu_pk.create_proof(inst, wit, rng) if s % 2 == 0 else \
+            oracle_prove(oracle_pk, inst, wit, rng)
+        items.append((inst, p))
+    assert gpu_pk.verify_batch(items)
+    assert gpu_pk.verify_batch(items[:1])  # m=1 path == single verifier
+    # tamper one proof's bytes -> whole batch rejected
+    bad = bytearray(items[2][1])
+    bad[100] ^= 1
+    assert not gpu_pk.verify_batch(items[:2] + [(items[2][0], bytes(bad))] + items[3:])
+    # swap one instance -> rejected
+    assert not gpu_pk.verify_batch(items[:3] + [(items[0][0], items[3][1])])
+    # truncated member -> rejected (structural)
+    assert not gpu_pk.verify_batch(items[:3] + [(items[3][0], items[3][1][:-32])])
+
+
+def test_batch_verify_agrees_with_single(gpu_pk):
+    """randomized agreement: for bundles of valid proofs the batch verdict
+    matches per-proof verification."""
+    gpu_pk.select_key(0)
+    items = []
+    for s in (7, 8, 9, 11, 12, 13):
+        inst = bytes([s]) + bytes(31)
+        p = gpu_pk.create_proof(inst, bytes([s + 1]) + bytes(31), bytes([s + 2]) + bytes(31))
+        assert gpu_pk.verify_proof(inst, p)
+        items.append((inst, p))
+    assert gpu_pk.verify_batch(items)
