@@ -70,7 +70,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--workload", choices=["proof", "msm", "ntt"], default="proof")
+    ap.add_argument("--workload", choices=["proof", "msm", "ntt", "verify"], default="proof")
     ap.add_argument("--streams", type=int, default=0,
                     help="concurrent proving contexts per GPU (proof workload); "
                          "0 = auto (6 single-rank, 2 when ranks share the host quota)")
@@ -131,6 +131,27 @@ def main():
             futs = [pool.submit(one_proof, ctxs[j], base + j) for j in range(C)]
             for f in futs:
                 f.result()
+    elif args.workload == "verify":
+        # batch verification: a step = one tg_verify_batch of
+        # a 6-proof bundle (a shielded ptx carries 2 compliance + 4 RL
+        # proofs - shielded_ptx.rs:137-153). Proofs are pre-generated
+        # (untimed); verification recomputes everything from the bytes, so
+        # reuse across steps caches nothing.
+        import pathlib
+
+        golden = pathlib.Path(REPO) / "tests" / "golden"
+        gpu.load_srs((golden / "params_15").read_bytes())
+        gpu.keygen((golden / "cs1.desc").read_bytes())
+        BUNDLE = 6
+        items = []
+        for i in range(BUNDLE):
+            inst = (SEED + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+            wit = (SEED + 77 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+            rng_s = (SEED + 99 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+            items.append((inst, gpu.create_proof(inst, wit, rng_s)))
+
+        def step():
+            assert gpu.verify_batch(items)
     elif args.workload == "msm":
         gpu.gen_bases(MSM_N, SEED)  # same base set on every rank
         scalars = gen_scalars(MSM_N, SEED + 1000 + rank)  # per-rank scalars
@@ -193,6 +214,16 @@ def main():
         workload_name = ("compliance_shaped_proof_k15 (CS1 stand-in: same shape/size as the "
                          "Action circuit — 10 advice, lookup, 12-col permutation, degree 9; "
                          "exact compliance witness fidelity is the round-2 item)")
+    elif args.workload == "verify":
+        acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
+        # one combined g-MSM (n=2^15) per batch regardless of bundle size
+        alg_bytes = 16 * (1 << 15) * 68
+        dom = ("msm_bucket_acc", acc_ms, acc_n, alg_bytes)
+        units_per_step = 6
+        unit = "proofs/s"
+        metric = "action_proofs_verified_per_sec"
+        workload_name = ("batch_verify_bundle6_k15 (one combined IPA check per "
+                         "6-proof bundle; CS1 Action-shaped circuit)")
     elif args.workload == "msm":
         acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
         # algorithmic bytes per k_bucket_acc launch (BASELINE.md config 2
@@ -278,6 +309,25 @@ def main():
                 "kind": "port",
                 "sample": "one CS1 proof via the oracle prover (OpenMP where parallel)",
             }
+        elif args.workload == "verify":
+            lib = oc.lib()
+            import pathlib
+
+            golden = pathlib.Path(REPO) / "tests" / "golden"
+            desc = (golden / "cs1.desc").read_bytes()
+            srs = (golden / "params_15").read_bytes()
+            assert lib.orc_prover_init(desc, len(desc), srs, len(srs)) in (0, 1)
+            inst, proof = items[0]
+            t0 = time.perf_counter()
+            assert lib.orc_verify_cs1(inst, proof, len(proof)) == 0
+            dt = time.perf_counter() - t0
+            cpu_baseline = {
+                "value": round(1.0 / dt, 4),
+                "unit": unit,
+                "cores": cores,
+                "kind": "port",
+                "sample": "one CS1 proof verification via the oracle (single check)",
+            }
         elif args.workload == "msm":
             nb = 1 << 17  # bounded sample (~10-30 s of CPU work)
             bases = oc.gen_bases(nb, SEED)
@@ -323,8 +373,8 @@ def main():
             "data": "synthetic",
             "config": {
                 "workload": workload_name,
-                "n_points": {"proof": 1 << 15, "msm": MSM_N, "ntt": 1 << NTT_K}[args.workload],
-                "window_bits": 16 if args.workload in ("proof", "msm") else None,
+                "n_points": {"proof": 1 << 15, "verify": 1 << 15, "msm": MSM_N, "ntt": 1 << NTT_K}[args.workload],
+                "window_bits": 16 if args.workload in ("proof", "verify", "msm") else None,
                 "parallelism": f"dp{n_gpus} (independent proofs/MSMs per GPU, no collective — SURVEY §8e)",
             },
             "roofline": roofline,
