@@ -132,6 +132,14 @@ int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* pro
  * malformed proof. */
 int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
                     const uint8_t* proofs, const size_t* proof_lens);
+/* batched Poseidon P128Pow5T3 ConstantLength<L> hashing over Fp (GPU
+ * witness synthesis — SURVEY §8f-2; replaces host halo2_gadgets poseidon
+ * hashing, utils.rs:40-48 / prf_nf utils.rs:37). msgs = n x L x 32B
+ * canonical reprs; out = n x 32B digests. One hash per GPU thread;
+ * parameters derived by tools/gen_poseidon.py (Grain LFSR), pinned against
+ * the oracle's independent derivation. */
+int tg_poseidon_hash(tg_ctx* ctx, const uint8_t* msgs, size_t n, int L,
+                     uint8_t* out);
 /* blake2b-256 of the generated advice matrix (witness-spec cross-check) */
 int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], uint8_t out[32]);

@@ -84,6 +84,10 @@ def load_library():
             ctypes.c_void_p, ctypes.c_size_t, ctypes.c_char_p, ctypes.c_char_p,
             ctypes.POINTER(ctypes.c_size_t),
         ]
+        lib.tg_poseidon_hash.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_int,
+            ctypes.c_char_p,
+        ]
         lib.tg_witness_hash.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
         ]
@@ -225,6 +229,13 @@ class TaigaGpu:
         if rc == -1 or rc <= -100:
             return False
         raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+
+    def poseidon_hash(self, msgs: bytes, n: int, L: int) -> bytes:
+        """Batched Poseidon P128Pow5T3 ConstantLength<L>: n messages of L
+        canonical 32B field reprs -> n 32B digests (one hash per thread)."""
+        out = ctypes.create_string_buffer(32 * n)
+        self._ck(self._lib.tg_poseidon_hash(self._h, msgs, n, L, out))
+        return out.raw
 
     def witness_hash(self, inst_seed: bytes, wit_seed: bytes) -> bytes:
         out = ctypes.create_string_buffer(32)

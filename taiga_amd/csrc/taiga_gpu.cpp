@@ -16,6 +16,7 @@
 
 #include "msm.hip"
 #include "ntt.hip"
+#include "poseidon.hip"
 #include "prover_impl.hpp"
 
 namespace taiga {
@@ -171,6 +172,7 @@ struct Ctx {
                                         // per tg_keygen call; slot id = index
 
   bool prof = false;
+  bool pos_ready = false;  // poseidon constants uploaded to __constant__
   ProfCounter prof_c[PROF_N];
 };
 
@@ -724,6 +726,48 @@ int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
     for (size_t i = 1; i < m; i++) rho[i] = rng.field<FpCfg>();
   }
   return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
+}
+
+/* batched Poseidon P128Pow5T3 ConstantLength<L> hashing (GPU witness
+ * synthesis; replaces host-side halo2_gadgets poseidon
+ * hashing — utils.rs:40-48): msgs = n x L x 32B canonical reprs,
+ * out = n x 32B canonical digests. One hash per GPU thread. */
+int tg_poseidon_hash(tg_ctx* ctx, const uint8_t* msgs, size_t n, int L,
+                     uint8_t* out) {
+  Ctx* c = (Ctx*)ctx;
+  if (!msgs || !out || n == 0 || L < 1 || L > 64) return TG_ERR_BADARG;
+  hipError_t e = hipSetDevice(c->device);
+  if (e != hipSuccess) return set_err(c, "hipSetDevice", e);
+  if (!c->pos_ready) {
+    if ((e = pos_upload_consts()) != hipSuccess)
+      return set_err(c, "poseidon consts", e);
+    c->pos_ready = true;
+  }
+  Fp* d_in = nullptr;
+  Fp* d_out = nullptr;
+  unsigned* d_err = nullptr;
+  size_t in_bytes = n * (size_t)L * 32;
+  if ((e = hipMalloc(&d_in, in_bytes)) != hipSuccess) return TG_ERR_NOMEM;
+  if ((e = hipMalloc(&d_out, n * 32)) != hipSuccess) { hipFree(d_in); return TG_ERR_NOMEM; }
+  if ((e = hipMalloc(&d_err, 4)) != hipSuccess) { hipFree(d_in); hipFree(d_out); return TG_ERR_NOMEM; }
+  hipMemsetAsync(d_err, 0, 4, c->stream);
+  hipMemcpyAsync(d_in, msgs, in_bytes, hipMemcpyHostToDevice, c->stream);
+  u64 grid = (n + 255) / 256;
+  if (grid > 16384) grid = 16384;
+  hipLaunchKernelGGL(k_poseidon_hash, dim3(grid), dim3(256), 0, c->stream,
+                     d_out, d_in, (u64)n, L, d_err);
+  unsigned h_err = 0;
+  hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, c->stream);
+  std::vector<uint8_t> h_out(n * 32);
+  hipMemcpyAsync(h_out.data(), d_out, n * 32, hipMemcpyDeviceToHost, c->stream);
+  e = hipStreamSynchronize(c->stream);
+  hipFree(d_in);
+  hipFree(d_out);
+  hipFree(d_err);
+  if (e != hipSuccess) return set_err(c, "poseidon", e);
+  if (h_err) return TG_ERR_ENCODING;
+  memcpy(out, h_out.data(), n * 32);
+  return TG_OK;
 }
 
 }  /* extern "C" */

@@ -196,3 +196,44 @@ def test_msm_skewed_scalars(gpu, params15):
         sc += (v if i > 100 else rng.randrange(pp.P)).to_bytes(32, "little")
     g = oc.decompress(oc.FQ, params15[4 : 4 + 32 * n])
     assert gpu.msm(bytes(sc), base_set=1) == oc.msm(oc.FQ, bytes(sc), g)
+
+
+# ---------------- Poseidon (GPU witness synthesis) ----------
+
+def test_poseidon_batch_parity(gpu):
+    """batched GPU Poseidon P128Pow5T3 vs the oracle's independent
+    implementation, several L and batch sizes including non-multiples of
+    the block size."""
+    import ctypes
+    import os
+
+    from conftest import REPO
+
+    orc = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+    rng = random.Random(4242)
+    for L, n in ((2, 1), (2, 1000), (1, 257), (9, 300), (5, 64)):
+        msgs = b"".join(
+            rng.randrange(pp.P).to_bytes(32, "little") for _ in range(n * L)
+        )
+        got = gpu.poseidon_hash(msgs, n, L)
+        for i in (0, n // 2, n - 1):
+            out = ctypes.create_string_buffer(32)
+            assert orc.orc_poseidon_hash(msgs[32 * L * i : 32 * L * (i + 1)], L, out) == 0
+            assert got[32 * i : 32 * i + 32] == out.raw
+    # full-batch check at L=2 against oracle for every element
+    n = 128
+    msgs = b"".join(rng.randrange(pp.P).to_bytes(32, "little") for _ in range(n * 2))
+    got = gpu.poseidon_hash(msgs, n, 2)
+    import ctypes as ct
+    for i in range(n):
+        out = ct.create_string_buffer(32)
+        assert orc.orc_poseidon_hash(msgs[64 * i : 64 * (i + 1)], 2, out) == 0
+        assert got[32 * i : 32 * i + 32] == out.raw
+
+
+def test_poseidon_rejects_noncanonical(gpu):
+    import taiga_amd
+
+    bad = pp.P.to_bytes(32, "little") + bytes(32)
+    with pytest.raises(taiga_amd.TaigaGpuError):
+        gpu.poseidon_hash(bad, 1, 2)
