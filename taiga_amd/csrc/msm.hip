@@ -23,19 +23,29 @@
 
 namespace taiga {
 
-// window size is a runtime parameter: c=16 for large MSMs (2^20 microbench),
-// c=12 for the prover's n=2^15 commits (11x smaller bucket space -> far
-// lower fixed cost per MSM). The result point is windowing-independent.
+// window size c and reduce-segment length are runtime parameters chosen by
+// MSM size: c=16 for large MSMs (2^20 microbench); c=13 with short segments
+// for the prover's n=2^15 commits — at n=2^15 the c=16 bucket space (16 x
+// 32768 buckets) dwarfs the point count and k_bucket_reduce's suffix-sum
+// walk over it was 44% of proof GPU time (profiles/
+// r01_final_proof_kernel_stats.csv); c=13 shrinks the bucket space 6.5x
+// (20 x 4096) for +25% bucket-accumulation adds. The result point is
+// windowing-independent. (c=12/seg=16 measured slower earlier: the reduce
+// kept its long segments and fell to 2.8k threads; segment length must
+// shrink with the bucket space.)
 constexpr int MSM_C_MAX = 16;
-constexpr int MSM_NWIN_MAX = 16;             // ceil(255/16)
+constexpr int MSM_NWIN_MAX = 16;             // ceil(255/16) (alloc worst case
+                                             // pairs with MSM_NBUCK_MAX;
+                                             // c=13 needs 20x4096 << 16x32768)
 constexpr int MSM_NBUCK_MAX = 1 << (MSM_C_MAX - 1);
-constexpr int MSM_SEG = 16;                  // buckets per reduction segment
+constexpr int MSM_SEG = 16;                  // seg len of the LARGE-c config
 struct MsmCfg {
-  int c, nwin, nbuck, nseg;
+  int c, nwin, nbuck, nseg, seg;
 };
 inline MsmCfg msm_cfg(long n) {
-  int c = 16;  /* measured: c=12 at n=2^15 costs more bucket adds than it saves (22 windows); keep 16 */
-  return MsmCfg{c, (255 + c - 1) / c, 1 << (c - 1), (1 << (c - 1)) / MSM_SEG};
+  if (n >= (1L << 18))
+    return MsmCfg{16, 16, 1 << 15, (1 << 15) / MSM_SEG, MSM_SEG};
+  return MsmCfg{13, 20, 1 << 12, (1 << 12) / 4, 4};
 }
 
 struct ScalarRepr {
@@ -290,15 +300,15 @@ __global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* bucket
   for (; t < ntot; t += (u64)gridDim.x * blockDim.x) {
     u64 w = t / cfg.nseg;
     u64 g = t % cfg.nseg;
-    const VestaJac* B = buckets + w * (u64)cfg.nbuck + g * MSM_SEG;
+    const VestaJac* B = buckets + w * (u64)cfg.nbuck + g * (u64)cfg.seg;
     VestaJac run = jac_identity<FqCfg>();
     VestaJac tot = jac_identity<FqCfg>();
-    for (int d = MSM_SEG - 1; d >= 0; d--) {
+    for (int d = cfg.seg - 1; d >= 0; d--) {
       run = jac_add(run, B[d]);
       tot = jac_add(tot, run);
     }
     // tot = sum (local_d+1) * B ; add base offset: (g*SEG) * run
-    u64 a = (u64)g * MSM_SEG;
+    u64 a = (u64)g * cfg.seg;
     // double-and-add small scalar a (< 2^15)
     VestaJac am = jac_identity<FqCfg>();
     VestaJac base = run;
