@@ -45,7 +45,20 @@ struct MsmCfg {
 inline MsmCfg msm_cfg(long n) {
   if (n >= (1L << 18))
     return MsmCfg{16, 16, 1 << 15, (1 << 15) / MSM_SEG, MSM_SEG};
-  return MsmCfg{13, 20, 1 << 12, (1 << 12) / 4, 4};
+  // perf-tuning overrides for the small-MSM config (A/B probes only;
+  // results are windowing-independent)
+  static int sc = [] {
+    const char* e = getenv("TG_MSM_SMALL_C");
+    int v = e ? atoi(e) : 13;
+    return (v >= 12 && v <= 16) ? v : 13; /* nwin <= 22 (d_dig alloc) */
+  }();
+  static int sseg = [] {
+    const char* e = getenv("TG_MSM_SMALL_SEG");
+    int v = e ? atoi(e) : 4;
+    return (v >= 1 && v <= 64) ? v : 4;
+  }();
+  int nbuck = 1 << (sc - 1);
+  return MsmCfg{sc, (255 + sc - 1) / sc, nbuck, nbuck / sseg, sseg};
 }
 
 struct ScalarRepr {
