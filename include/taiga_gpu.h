@@ -123,6 +123,11 @@ int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* adv
  * SingleVerifier strategy — proof.rs:45-54). Returns TG_OK iff valid. */
 int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* proof,
                     size_t proof_len);
+/* raw-instance verification: instance = n_instance_rows x 32B canonical
+ * reprs (the drop-in shape of plonk::verify_proof's instance slices —
+ * proof.rs:45-54). */
+int tg_verify_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* proof,
+                        size_t proof_len);
 /* batch verification (halo2's BatchVerifier accumulation for
  * a bundle's proofs — shielded_ptx.rs:137-153): all m final IPA checks
  * collapse into one randomly-weighted combined check sharing a single

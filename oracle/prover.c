@@ -941,8 +941,11 @@ typedef struct {
     fd_limbs blind;
 } MPoly;
 
-int orc_prove(Pk* pk, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
-              const uint8_t rng_seed[32], uint8_t** out, size_t* out_len) {
+/* core prover over explicit instance column + advice matrix (Mont form;
+ * advice rows >= usable are overwritten by blinding). The seeded wrapper
+ * below and the raw-bytes entry both funnel here. */
+static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_in,
+                          const uint8_t rng_seed[32], uint8_t** out, size_t* out_len) {
     Desc* d = pk->d;
     long n = d->n, ext_n = d->ext_n, u = d->usable;
     int nchunks = (d->n_perm + d->chunk_len - 1) / d->chunk_len;
@@ -958,7 +961,7 @@ int orc_prove(Pk* pk, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
 
     /* 1. instance */
     fd_limbs* inst_lag = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
-    orc_cs1_instance(d, inst_seed, inst_lag);
+    memcpy(inst_lag, inst_in, sizeof(fd_limbs) * (size_t)n);
     pt_aff inst_commit;
     commit_msm(&inst_commit, inst_lag, n, pk->gl, one, pk);
     orc_ts_common_point(&ts, &inst_commit);
@@ -976,7 +979,8 @@ int orc_prove(Pk* pk, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
         advice_coeff[c] = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
         advice_ext[c] = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)ext_n);
     }
-    orc_cs1_witness(d, wit_seed, inst_lag, advice_lag);
+    for (int c = 0; c < d->n_advice; c++)
+        memcpy(advice_lag[c], adv_in[c], sizeof(fd_limbs) * (size_t)n);
     /* blinding rows then blinds (DESIGN.md draw order) */
     for (int c = 0; c < d->n_advice; c++)
         for (long i = u; i < n; i++) prng_field(&rng, advice_lag[c][i]);
@@ -1968,7 +1972,8 @@ static void pt_muladd(pt_aff* acc /* in/out */, const fd_limbs s_mont, const pt_
     pt_to_aff(acc, &r, FQ);
 }
 
-int orc_verify(Pk* pk, const uint8_t inst_seed[32], const uint8_t* proof, size_t proof_len) {
+static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof,
+                           size_t proof_len) {
     Desc* d = pk->d;
     long n = d->n;
     int nchunks = (d->n_perm + d->chunk_len - 1) / d->chunk_len;
@@ -1983,7 +1988,7 @@ int orc_verify(Pk* pk, const uint8_t inst_seed[32], const uint8_t* proof, size_t
 
     /* instance commitment recomputed from public input */
     fd_limbs* inst_lag = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
-    orc_cs1_instance(d, inst_seed, inst_lag);
+    memcpy(inst_lag, inst_in, sizeof(fd_limbs) * (size_t)n);
     pt_aff inst_commit;
     commit_msm(&inst_commit, inst_lag, n, pk->gl, one, pk);
     orc_ts_common_point(&ts, &inst_commit);
@@ -2475,6 +2480,89 @@ pt_aff* orc_pk_fixed_commits(Pk* pk) {
 static Pk* g_pk = NULL;
 static uint8_t* g_desc_copy = NULL;
 static uint8_t* g_srs_copy = NULL;
+
+int orc_verify(Pk* pk, const uint8_t inst_seed[32], const uint8_t* proof, size_t proof_len) {
+    Desc* d = pk->d;
+    fd_limbs* inst = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)d->n);
+    orc_cs1_instance(d, inst_seed, inst);
+    int rc = orc_verify_core(pk, inst, proof, proof_len);
+    free(inst);
+    return rc;
+}
+
+int orc_prove(Pk* pk, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
+              const uint8_t rng_seed[32], uint8_t** out, size_t* out_len) {
+    Desc* d = pk->d;
+    long n = d->n;
+    fd_limbs* inst = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
+    orc_cs1_instance(d, inst_seed, inst);
+    fd_limbs* adv[16];
+    for (int c = 0; c < d->n_advice; c++)
+        adv[c] = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
+    orc_cs1_witness(d, wit_seed, inst, adv);
+    int rc = orc_prove_core(pk, inst, adv, rng_seed, out, out_len);
+    free(inst);
+    for (int c = 0; c < d->n_advice; c++) free(adv[c]);
+    return rc;
+}
+
+/* canonical instance rows -> Mont instance column; returns 0 or -3 */
+static int orc_inst_from_raw(const Desc* d, const uint8_t* inst_bytes, fd_limbs* inst) {
+    for (long i = 0; i < d->n; i++) fd_zero(inst[i]);
+    for (int r = 0; r < d->n_instance_rows; r++)
+        if (fd_from_bytes(inst[r], inst_bytes + 32 * r, FP)) return -3;
+    return 0;
+}
+
+/* raw-witness prove (mirrors tg_create_proof_raw): instance =
+ * n_instance_rows x 32B canonical; advice = n_advice x n x 32B canonical
+ * column-major. Writes the proof into out (cap bytes); returns length. */
+long orc_prove_raw(const uint8_t* inst_bytes, const uint8_t* adv_bytes,
+                   const uint8_t rng_seed[32], uint8_t* out, long cap) {
+    if (!g_pk) return -1;
+    Desc* d = g_pk->d;
+    long n = d->n;
+    fd_limbs* inst = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
+    if (orc_inst_from_raw(d, inst_bytes, inst)) { free(inst); return -3; }
+    fd_limbs* adv[16];
+    int bad = 0;
+    for (int c = 0; c < d->n_advice; c++) {
+        adv[c] = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)n);
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static) reduction(| : bad)
+#endif
+        for (long i = 0; i < n; i++)
+            if (fd_from_bytes(adv[c][i], adv_bytes + ((size_t)c * n + i) * 32, FP))
+                bad |= 1;
+    }
+    long ret = -3;
+    if (!bad) {
+        uint8_t* proof = NULL;
+        size_t plen = 0;
+        int rc = orc_prove_core(g_pk, inst, adv, rng_seed, &proof, &plen);
+        if (rc == 0 && (long)plen <= cap) {
+            memcpy(out, proof, plen);
+            ret = (long)plen;
+        } else {
+            ret = rc ? rc : -2;
+        }
+        free(proof);
+    }
+    free(inst);
+    for (int c = 0; c < d->n_advice; c++) free(adv[c]);
+    return ret;
+}
+
+/* raw-instance verify (mirrors tg_verify_proof_raw) */
+int orc_verify_raw(const uint8_t* inst_bytes, const uint8_t* proof, long plen) {
+    if (!g_pk) return -1;
+    Desc* d = g_pk->d;
+    fd_limbs* inst = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (size_t)d->n);
+    if (orc_inst_from_raw(d, inst_bytes, inst)) { free(inst); return -3; }
+    int rc = orc_verify_core(g_pk, inst, proof, (size_t)plen);
+    free(inst);
+    return rc;
+}
 
 int orc_prover_init(const uint8_t* desc, long desc_len, const uint8_t* srs, long srs_len) {
     if (g_pk) return 1; /* already initialized (desc assumed identical) */

@@ -80,6 +80,9 @@ def load_library():
         lib.tg_verify_proof.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
         ]
+        lib.tg_verify_proof_raw.argtypes = [
+            ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
+        ]
         lib.tg_verify_batch.argtypes = [
             ctypes.c_void_p, ctypes.c_size_t, ctypes.c_char_p, ctypes.c_char_p,
             ctypes.POINTER(ctypes.c_size_t),
@@ -212,6 +215,14 @@ class TaigaGpu:
         if rc == 0:
             return True
         # -1 = final check failed; -1xx = malformed/truncated transcript
+        if rc == -1 or rc <= -100:
+            return False
+        raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+
+    def verify_proof_raw(self, instance: bytes, proof: bytes) -> bool:
+        rc = self._lib.tg_verify_proof_raw(self._h, instance, proof, len(proof))
+        if rc == 0:
+            return True
         if rc == -1 or rc <= -100:
             return False
         raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())

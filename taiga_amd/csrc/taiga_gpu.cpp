@@ -690,6 +690,17 @@ int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* pro
   return pverify(c, *c->ppk, inst_seed, proof, proof_len);
 }
 
+/* raw-instance verification: instance = n_instance_rows x 32B canonical
+ * reprs (the drop-in shape of plonk::verify_proof's instance slices —
+ * proof.rs:45-54). */
+int tg_verify_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* proof,
+                        size_t proof_len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  if (!instance) return TG_ERR_BADARG;
+  return pverify_raw(c, *c->ppk, instance, proof, proof_len);
+}
+
 /* batch verification (the halo2 BatchVerifier / Guard
  * accumulation pattern behind proof.rs:45-54): each proof contributes a
  * final-check guard; guards are combined with random weights so that all
@@ -706,7 +717,9 @@ int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
   std::vector<PVGuard> gds(m);
   size_t off = 0;
   for (size_t i = 0; i < m; i++) {
-    int rc = pverify_guard(c, *c->ppk, inst_seeds + 32 * i, proofs + off,
+    std::vector<Fp> inst_lag;
+    cs1_instance(c->ppk->d, inst_seeds + 32 * i, inst_lag);
+    int rc = pverify_guard(c, *c->ppk, inst_lag, proofs + off,
                            proof_lens[i], gds[i]);
     if (rc != 0) return rc;
     off += proof_lens[i];
