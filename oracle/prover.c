@@ -2564,6 +2564,68 @@ int orc_verify_raw(const uint8_t* inst_bytes, const uint8_t* proof, long plen) {
     return rc;
 }
 
+static void desc_free(Desc* d) {
+    if (!d) return;
+    free(d->consts);
+    free(d->advice_q);
+    free(d->fixed_q);
+    free(d->instance_q);
+    free(d->perm_cols);
+    for (int i = 0; i < d->n_gates; i++) free(d->gates[i].ops);
+    free(d->gates);
+    for (int i = 0; i < d->n_lookups; i++) {
+        for (uint32_t j = 0; j < d->lookups[i].n_in; j++) free(d->lookups[i].in[j].ops);
+        for (uint32_t j = 0; j < d->lookups[i].n_tab; j++) free(d->lookups[i].tab[j].ops);
+        free(d->lookups[i].in);
+        free(d->lookups[i].tab);
+    }
+    free(d->lookups);
+    free(d->sigma_map);
+    for (int c = 0; c < d->n_fixed; c++) free(d->fixed_lag[c]);
+    free(d->fixed_lag);
+    free(d);
+}
+
+static void pk_free(Pk* pk) {
+    if (!pk) return;
+    Desc* d = pk->d;
+    for (int c = 0; c < d->n_fixed; c++) {
+        free(pk->fixed_coeff[c]);
+        free(pk->fixed_ext[c]);
+    }
+    free(pk->fixed_coeff);
+    free(pk->fixed_ext);
+    for (int j = 0; j < d->n_perm; j++) {
+        free(pk->sigma_lag[j]);
+        free(pk->sigma_coeff[j]);
+        free(pk->sigma_ext[j]);
+    }
+    free(pk->sigma_lag);
+    free(pk->sigma_coeff);
+    free(pk->sigma_ext);
+    free(pk->l0_ext);
+    free(pk->llast_ext);
+    free(pk->lactive_ext);
+    free(pk->t_inv_ext);
+    free(pk->x_ext);
+    free(pk->g);
+    free(pk->gl);
+    free(pk->sigma_commits);
+    desc_free(d);
+    free(pk);
+}
+
+/* drop the cached proving key so a different circuit desc can be loaded
+ * (test infrastructure for multi-circuit parity fuzzing) */
+void orc_prover_reset(void) {
+    pk_free(g_pk);
+    g_pk = NULL;
+    free(g_desc_copy);
+    g_desc_copy = NULL;
+    free(g_srs_copy);
+    g_srs_copy = NULL;
+}
+
 int orc_prover_init(const uint8_t* desc, long desc_len, const uint8_t* srs, long srs_len) {
     if (g_pk) return 1; /* already initialized (desc assumed identical) */
     g_desc_copy = (uint8_t*)xmalloc((size_t)desc_len);
