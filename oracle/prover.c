@@ -2459,10 +2459,13 @@ static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof
     return 0;
 }
 
-/* fixed commitments cache */
+/* fixed commitments cache (invalidated by orc_prover_reset: a new Pk can
+ * be allocated at the old one's address) */
+static pt_aff* g_fixed_cms_cache = NULL;
+static Pk* g_fixed_cms_pk = NULL;
 pt_aff* orc_pk_fixed_commits(Pk* pk) {
-    static pt_aff* cache = NULL;
-    static Pk* cache_pk = NULL;
+#define cache g_fixed_cms_cache
+#define cache_pk g_fixed_cms_pk
     if (cache_pk == pk && cache) return cache;
     Desc* d = pk->d;
     pt_aff* cms = (pt_aff*)xmalloc(sizeof(pt_aff) * (size_t)d->n_fixed);
@@ -2470,9 +2473,12 @@ pt_aff* orc_pk_fixed_commits(Pk* pk) {
     fd_one_mont(one, FP);
     for (int c = 0; c < d->n_fixed; c++)
         commit_msm(&cms[c], d->fixed_lag[c], d->n, pk->gl, one, pk);
+    free(cache);
     cache = cms;
     cache_pk = pk;
     return cms;
+#undef cache
+#undef cache_pk
 }
 
 /* ---------------- ctypes-facing API ---------------- */
@@ -2618,6 +2624,9 @@ static void pk_free(Pk* pk) {
 /* drop the cached proving key so a different circuit desc can be loaded
  * (test infrastructure for multi-circuit parity fuzzing) */
 void orc_prover_reset(void) {
+    free(g_fixed_cms_cache);
+    g_fixed_cms_cache = NULL;
+    g_fixed_cms_pk = NULL;
     pk_free(g_pk);
     g_pk = NULL;
     free(g_desc_copy);
