@@ -137,15 +137,18 @@ def gen(seed):
         copy((0, 2), r1, (0, 2), r2)
     chunk_len = rng.randint(3, 7)
 
-    # ---- lookups: input [q_lk * a1], table [f_table]; a1 values on active
-    # rows drawn from the table's value set (0 in the set for idle rows)
+    # ---- lookups: input [q_lk * a1], table [f_table]. All lookups share
+    # one value pool S (their tables hold the same SET in different row
+    # arrangements) so overlapping active regions on the shared input
+    # column a1 stay members of every table; 0 in S covers idle rows.
     lookups = []
+    tset = [0] + [rng.randrange(P) for _ in range(63)]
     for l in range(n_lookups):
         qcol = n_gates + 2 * l
         tcol = n_gates + 2 * l + 1
-        tset = [0] + [rng.randrange(P) for _ in range(63)]
+        off = rng.randrange(64)
         for i in range(N):
-            fixed[tcol][i] = tset[i % 64]
+            fixed[tcol][i] = tset[(i + off) % 64]
         lo = rng.randrange(0, usable // 2)
         hi = rng.randrange(lo + 1, usable - 1)
         for i in range(lo, hi):
