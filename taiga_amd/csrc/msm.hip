@@ -45,19 +45,27 @@ struct MsmCfg {
 inline MsmCfg msm_cfg(long n) {
   if (n >= (1L << 18))
     return MsmCfg{16, 16, 1 << 15, (1 << 15) / MSM_SEG, MSM_SEG};
-  // perf-tuning overrides for the small-MSM config (A/B probes only;
-  // results are windowing-independent)
-  static int sc = [] {
+  // below 2^18 the bucket space must track the MSM size (c ~ log2 n - 2,
+  // floored at 8 to keep nwin <= 32 for the d_dig/d_wsums allocs): the prover's
+  // n=2^15 commits take c=13, the IPA's geometrically shrinking rounds
+  // smaller windows still. Reduce segments sized so nseg <= 1024 per
+  // window. TG_MSM_SMALL_C/SEG override for A/B probes only (results are
+  // windowing-independent).
+  static int sc_env = [] {
     const char* e = getenv("TG_MSM_SMALL_C");
-    int v = e ? atoi(e) : 13;
-    return (v >= 12 && v <= 16) ? v : 13; /* nwin <= 22 (d_dig alloc) */
+    int v = e ? atoi(e) : 0;
+    return (v >= 8 && v <= 16) ? v : 0; /* nwin <= 32 (d_dig/d_wsums allocs) */
   }();
-  static int sseg = [] {
+  static int sseg_env = [] {
     const char* e = getenv("TG_MSM_SMALL_SEG");
-    int v = e ? atoi(e) : 4;
-    return (v >= 1 && v <= 64) ? v : 4;
+    int v = e ? atoi(e) : 0;
+    return (v >= 1 && v <= 64) ? v : 0;
   }();
+  int lg = 0;
+  while ((1L << (lg + 1)) <= n) lg++;
+  int sc = sc_env ? sc_env : (lg - 2 < 8 ? 8 : (lg - 2 > 13 ? 13 : lg - 2));
   int nbuck = 1 << (sc - 1);
+  int sseg = sseg_env ? sseg_env : (nbuck / 1024 > 0 ? nbuck / 1024 : 1);
   return MsmCfg{sc, (255 + sc - 1) / sc, nbuck, nbuck / sseg, sseg};
 }
 
@@ -405,13 +413,13 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n_total, u64 batch = 1) {
   TGW_FREE(w.d_bsum) TGW_FREE(w.d_sorted) TGW_FREE(w.d_buckets) TGW_FREE(w.d_partials)
   TGW_FREE(w.d_wsums) TGW_FREE(w.d_big)
 #undef TGW_FREE
-  if ((e = hipMalloc(&w.d_dig, n_total * 22 * 4)) != hipSuccess) return e;  /* nwin<=22 (c=12) */
+  if ((e = hipMalloc(&w.d_dig, n_total * 32 * 4)) != hipSuccess) return e;  /* nwin<=32 (c=8) */
   if ((e = hipMalloc(&w.d_hist, m * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_off, m * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_end, m * 4)) != hipSuccess) return e;
   u64 nb1 = (m + 511) / 512;
   if ((e = hipMalloc(&w.d_bsum, (nb1 + (nb1 + 511) / 512 + 1) * 4)) != hipSuccess) return e;
-  if ((e = hipMalloc(&w.d_sorted, n_total * 22 * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_sorted, n_total * 32 * 4)) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_buckets, m * sizeof(VestaJac))) != hipSuccess) return e;
   if ((e = hipMalloc(&w.d_partials, (u64)MSM_NWIN_MAX * 2 * batch * MSM_NBUCK_MAX /
                                         MSM_SEG * sizeof(VestaJac))) != hipSuccess)
