@@ -500,7 +500,7 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
                          c->msm.d_partials, c->msm.d_wsums, cfg);
     }
   }
-  VestaJac wsums[22];
+  VestaJac wsums[32]; /* nwin <= 32 (c >= 8) */
   hipMemcpyAsync(wsums, c->msm.d_wsums, sizeof(VestaJac) * cfg.nwin,
                  hipMemcpyDeviceToHost, c->stream);
   hipError_t es = hipStreamSynchronize(c->stream);
