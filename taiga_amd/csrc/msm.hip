@@ -342,6 +342,17 @@ __global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* bucket
   }
 }
 
+// minimum total window count (workgroups) for the fused reduce below to
+// beat the split kernels: a B=1 MSM gives it only ~20 workgroups, which
+// strands most of the chip. TG_MSM_FUSED_MIN overrides for A/B probes.
+static inline int msm_fused_min() {
+  static int v = [] {
+    const char* e = getenv("TG_MSM_FUSED_MIN");
+    return e ? atoi(e) : 64;
+  }();
+  return v;
+}
+
 // fused segment-reduce + window-sum for the SMALL-MSM configs
 // (nseg <= 1024): workgroup w covers window w entirely — each thread
 // suffix-sums up to 4 segments, an LDS tree folds the 256 thread

@@ -488,7 +488,7 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
                            c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
       }
     }
-    if (cfg.nseg <= 1024) {
+    if (cfg.nseg <= 1024 && cfg.nwin >= msm_fused_min()) {
       ProfScope p(c, P_MSM_REDUCE);
       hipLaunchKernelGGL(k_bucket_reduce_wsum, dim3(cfg.nwin), dim3(256), 0,
                          c->stream, c->msm.d_buckets, c->msm.d_wsums, cfg);
