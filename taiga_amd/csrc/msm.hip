@@ -342,56 +342,6 @@ __global__ void __launch_bounds__(256, 1) k_bucket_reduce(const VestaJac* bucket
   }
 }
 
-// minimum total window count (workgroups) for the fused reduce below to
-// beat the split kernels: a B=1 MSM gives it only ~20 workgroups, which
-// strands most of the chip. TG_MSM_FUSED_MIN overrides for A/B probes.
-static inline int msm_fused_min() {
-  static int v = [] {
-    const char* e = getenv("TG_MSM_FUSED_MIN");
-    return e ? atoi(e) : 64;
-  }();
-  return v;
-}
-
-// fused segment-reduce + window-sum for the SMALL-MSM configs
-// (nseg <= 1024): workgroup w covers window w entirely — each thread
-// suffix-sums up to 4 segments, an LDS tree folds the 256 thread
-// accumulators, and wsums[w] is written directly. No partials round-trip
-// through HBM and one dispatch instead of two. (The c=16 config keeps the
-// split kernels: 16 workgroups would strand 94% of the chip.)
-__global__ void __launch_bounds__(256, 1) k_bucket_reduce_wsum(const VestaJac* buckets,
-                                                               VestaJac* wsums, MsmCfg cfg) {
-  __shared__ VestaJac lds[256];
-  u64 w = blockIdx.x;  // window index across the batch
-  int t = threadIdx.x;
-  VestaJac acc = jac_identity<FqCfg>();
-  for (int g = t; g < cfg.nseg; g += 256) {
-    const VestaJac* B = buckets + w * (u64)cfg.nbuck + (u64)g * cfg.seg;
-    VestaJac run = jac_identity<FqCfg>();
-    VestaJac tot = jac_identity<FqCfg>();
-    for (int d = cfg.seg - 1; d >= 0; d--) {
-      run = jac_add(run, B[d]);
-      tot = jac_add(tot, run);
-    }
-    u64 a = (u64)g * cfg.seg;
-    VestaJac am = jac_identity<FqCfg>();
-    VestaJac base = run;
-    while (a) {
-      if (a & 1) am = jac_add(am, base);
-      base = jac_dbl(base);
-      a >>= 1;
-    }
-    acc = jac_add(acc, jac_add(tot, am));
-  }
-  lds[t] = acc;
-  __syncthreads();
-  for (int off = 128; off >= 1; off >>= 1) {
-    if (t < off) lds[t] = jac_add(lds[t], lds[t + off]);
-    __syncthreads();
-  }
-  if (t == 0) wsums[w] = lds[0];
-}
-
 // per-window tree reduce: block w reduces its MSM_NSEG partials to one
 // Jacobian window sum (the 16 window sums go to the host shim, which does
 // the O(1) 240-doubling Horner combine with the same TG_HD primitives —

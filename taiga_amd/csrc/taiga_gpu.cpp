@@ -488,17 +488,13 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
                            c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
       }
     }
-    if (cfg.nseg <= 1024 && cfg.nwin >= msm_fused_min()) {
+    {
       ProfScope p(c, P_MSM_REDUCE);
-      hipLaunchKernelGGL(k_bucket_reduce_wsum, dim3(cfg.nwin), dim3(256), 0,
-                         c->stream, c->msm.d_buckets, c->msm.d_wsums, cfg);
-    } else {
-      {
-        ProfScope p(c, P_MSM_REDUCE);
-        hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)cfg.nwin * cfg.nseg)),
-                           dim3(256), 0, c->stream, c->msm.d_buckets, c->msm.d_partials,
-                           (u64)cfg.nwin, cfg);
-      }
+      hipLaunchKernelGGL(k_bucket_reduce, dim3(msm_grid((u64)cfg.nwin * cfg.nseg)),
+                         dim3(256), 0, c->stream, c->msm.d_buckets, c->msm.d_partials,
+                         (u64)cfg.nwin, cfg);
+    }
+    {
       ProfScope p(c, P_MSM_WSUM);
       hipLaunchKernelGGL(k_wsum, dim3(cfg.nwin), dim3(256), 0, c->stream,
                          c->msm.d_partials, c->msm.d_wsums, cfg);
