@@ -149,6 +149,33 @@ int tg_poseidon_hash(tg_ctx* ctx, const uint8_t* msgs, size_t n, int L,
 int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], uint8_t out[32]);
 
+/* ---- binding signatures + transaction digest (host-side wire layer,
+ *      SURVEY §8f-4) ----
+ * RedDSA over Pallas as instantiated by TaigaBinding
+ * (binding_signature.rs:23-31): H* = BLAKE2b-512 with personalization
+ * "Taiga_RedPallasH" wide-reduced to the Pallas scalar field; signature =
+ * 32B compressed R ‖ 32B LE scalar S; sign nonce = H*(T ‖ vk ‖ msg) with
+ * T = 80 ChaCha20-DRBG bytes from rng_seed. Ctx-free host calls. The
+ * basepoint is the Pallas generator pending the sinsemilla group-hash
+ * chain for RESOURCE_COMMIT_DOMAIN.R() (constant.rs:160).
+ * Scalars are 32B LE canonical mod the Pallas group order. */
+int tg_binding_vk(const uint8_t sk[32], uint8_t vk_out[32]);
+int tg_delta_commit(const uint8_t r[32], uint8_t cv_out[32]);
+int tg_binding_sign(const uint8_t sk[32], const uint8_t* msg, size_t msg_len,
+                    const uint8_t rng_seed[32], uint8_t sig_out[64]);
+/* TG_OK iff the signature verifies */
+int tg_binding_verify(const uint8_t vk[32], const uint8_t* msg, size_t msg_len,
+                      const uint8_t sig[64]);
+/* binding vk = sum of a bundle's delta commitments (transaction.rs:99-114) */
+int tg_binding_vk_from_deltas(const uint8_t* deltas, size_t n, uint8_t vk_out[32]);
+/* Transaction::digest (transaction.rs:116-158): BLAKE2b-256
+ * "TxBindingSigHash" over nullifiers ‖ output cms ‖ delta commitments ‖
+ * anchors (32B each; append the transparent bundle's streams after the
+ * shielded ones). */
+int tg_tx_digest(const uint8_t* nfs, size_t n_nf, const uint8_t* cms, size_t n_cm,
+                 const uint8_t* deltas, size_t n_delta, const uint8_t* anchors,
+                 size_t n_anchor, uint8_t out[32]);
+
 /* ---- kernel profiling (HIP events on the ctx stream) ----
  * names: "msm_digits", "msm_scan", "msm_scatter", "msm_bucket_acc",
  *        "msm_reduce", "msm_wsum", "ntt_stage", "ntt_fused", "ntt_bitrev",

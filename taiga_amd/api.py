@@ -94,6 +94,19 @@ def load_library():
         lib.tg_witness_hash.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_char_p,
         ]
+        lib.tg_binding_vk.argtypes = [ctypes.c_char_p, ctypes.c_char_p]
+        lib.tg_delta_commit.argtypes = [ctypes.c_char_p, ctypes.c_char_p]
+        lib.tg_binding_sign.argtypes = [
+            ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
+            ctypes.c_char_p,
+        ]
+        lib.tg_binding_verify.argtypes = [
+            ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
+        ]
+        lib.tg_binding_vk_from_deltas.argtypes = [
+            ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
+        ]
+        lib.tg_tx_digest.argtypes = [ctypes.c_char_p, ctypes.c_size_t] * 4 + [ctypes.c_char_p]
         lib.tg_prof_enable.argtypes = [ctypes.c_void_p, ctypes.c_int]
         lib.tg_prof_enable.restype = None
         lib.tg_prof_reset.argtypes = [ctypes.c_void_p]
@@ -105,6 +118,41 @@ def load_library():
         lib.tg_synchronize.argtypes = [ctypes.c_void_p]
         _lib = lib
     return _lib
+
+
+def binding_sign(sk: bytes, msg: bytes, rng_seed: bytes) -> bytes:
+    """RedDSA binding signature (TaigaBinding instantiation) — ctx-free
+    host call (wire layer)."""
+    lib = load_library()
+    sig = ctypes.create_string_buffer(64)
+    rc = lib.tg_binding_sign(sk, msg, len(msg), rng_seed, sig)
+    if rc != 0:
+        raise TaigaGpuError(rc)
+    return sig.raw
+
+
+def binding_verify(vk: bytes, msg: bytes, sig: bytes) -> bool:
+    return load_library().tg_binding_verify(vk, msg, len(msg), sig) == 0
+
+
+def binding_vk(sk: bytes) -> bytes:
+    lib = load_library()
+    out = ctypes.create_string_buffer(32)
+    rc = lib.tg_binding_vk(sk, out)
+    if rc != 0:
+        raise TaigaGpuError(rc)
+    return out.raw
+
+
+def tx_digest(nullifiers, output_cms, delta_cms, anchors) -> bytes:
+    """Transaction::digest layout (transaction.rs:116-158)."""
+    lib = load_library()
+    out = ctypes.create_string_buffer(32)
+    lib.tg_tx_digest(b"".join(nullifiers), len(nullifiers),
+                     b"".join(output_cms), len(output_cms),
+                     b"".join(delta_cms), len(delta_cms),
+                     b"".join(anchors), len(anchors), out)
+    return out.raw
 
 
 def device_count():

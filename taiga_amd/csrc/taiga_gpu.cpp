@@ -17,6 +17,7 @@
 #include "msm.hip"
 #include "ntt.hip"
 #include "poseidon.hip"
+#include "binding_sig.hpp"
 #include "prover_impl.hpp"
 
 namespace taiga {
@@ -739,6 +740,45 @@ int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
     for (size_t i = 1; i < m; i++) rho[i] = rng.field<FpCfg>();
   }
   return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
+}
+
+/* ---- binding signatures + transaction digest (host-side wire layer,
+ * SURVEY §8f-4; binding_signature.rs / transaction.rs:116-158) ----
+ * RedDSA over Pallas, H* = BLAKE2b-512("Taiga_RedPallasH"), ctx-free.
+ * See binding_sig.hpp for the basepoint pin status. */
+int tg_binding_vk(const uint8_t sk[32], uint8_t vk_out[32]) {
+  return bs_derive_vk(vk_out, sk) ? TG_ERR_ENCODING : TG_OK;
+}
+
+int tg_delta_commit(const uint8_t r[32], uint8_t cv_out[32]) {
+  Fq s;
+  if (!bs_scalar_from_bytes(s, r)) return TG_ERR_ENCODING;
+  pallas_compress(cv_out, pallas_mul(pallas_basepoint(), s));
+  return TG_OK;
+}
+
+int tg_binding_sign(const uint8_t sk[32], const uint8_t* msg, size_t msg_len,
+                    const uint8_t rng_seed[32], uint8_t sig_out[64]) {
+  if (!msg && msg_len) return TG_ERR_BADARG;
+  return bs_sign(sig_out, sk, msg, msg_len, rng_seed) ? TG_ERR_ENCODING : TG_OK;
+}
+
+int tg_binding_verify(const uint8_t vk[32], const uint8_t* msg, size_t msg_len,
+                      const uint8_t sig[64]) {
+  if (!msg && msg_len) return TG_ERR_BADARG;
+  return bs_verify(vk, msg, msg_len, sig) ? -1 : TG_OK;
+}
+
+int tg_binding_vk_from_deltas(const uint8_t* deltas, size_t n, uint8_t vk_out[32]) {
+  if (n && !deltas) return TG_ERR_BADARG;
+  return bs_vk_from_deltas(vk_out, deltas, n) ? TG_ERR_ENCODING : TG_OK;
+}
+
+int tg_tx_digest(const uint8_t* nfs, size_t n_nf, const uint8_t* cms, size_t n_cm,
+                 const uint8_t* deltas, size_t n_delta, const uint8_t* anchors,
+                 size_t n_anchor, uint8_t out[32]) {
+  bs_tx_digest(out, nfs, n_nf, cms, n_cm, deltas, n_delta, anchors, n_anchor);
+  return TG_OK;
 }
 
 /* batched Poseidon P128Pow5T3 ConstantLength<L> hashing (GPU witness
