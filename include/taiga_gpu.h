@@ -137,6 +137,10 @@ int tg_verify_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* pro
  * malformed proof. */
 int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
                     const uint8_t* proofs, const size_t* proof_lens);
+/* raw-instance batch verification (the real ptx-bundle shape): instances =
+ * m concatenated n_instance_rows x 32B blocks for the active key. */
+int tg_verify_batch_raw(tg_ctx* ctx, size_t m, const uint8_t* instances,
+                        const uint8_t* proofs, const size_t* proof_lens);
 /* batched Poseidon P128Pow5T3 ConstantLength<L> hashing over Fp (GPU
  * witness synthesis — SURVEY §8f-2; replaces host halo2_gadgets poseidon
  * hashing, utils.rs:40-48 / prf_nf utils.rs:37). msgs = n x L x 32B

@@ -87,6 +87,10 @@ def load_library():
             ctypes.c_void_p, ctypes.c_size_t, ctypes.c_char_p, ctypes.c_char_p,
             ctypes.POINTER(ctypes.c_size_t),
         ]
+        lib.tg_verify_batch_raw.argtypes = [
+            ctypes.c_void_p, ctypes.c_size_t, ctypes.c_char_p, ctypes.c_char_p,
+            ctypes.POINTER(ctypes.c_size_t),
+        ]
         lib.tg_poseidon_hash.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_int,
             ctypes.c_char_p,
@@ -286,6 +290,20 @@ class TaigaGpu:
         if rc == 0:
             return True
         if rc == -1 or rc <= -100:
+            return False
+        raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+
+    def verify_batch_raw(self, items) -> bool:
+        """Batch-verify [(instance_rows_bytes, proof), ...] (raw instances,
+        active key) in one combined IPA check."""
+        m = len(items)
+        insts = b"".join(i for i, _ in items)
+        proofs = b"".join(p for _, p in items)
+        lens = (ctypes.c_size_t * m)(*[len(p) for _, p in items])
+        rc = self._lib.tg_verify_batch_raw(self._h, m, insts, proofs, lens)
+        if rc == 0:
+            return True
+        if rc == -1 or rc <= -100 or rc == -3:
             return False
         raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
 

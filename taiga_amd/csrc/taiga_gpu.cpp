@@ -742,6 +742,42 @@ int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
   return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
 }
 
+/* raw-instance batch verification: the ptx-bundle shape (§8f-3 + §8f-4
+ * composed): instances = concatenated n_instance_rows x 32B blocks, one
+ * per proof, for the ACTIVE key. Same combined-check semantics as
+ * tg_verify_batch. */
+int tg_verify_batch_raw(tg_ctx* ctx, size_t m, const uint8_t* instances,
+                        const uint8_t* proofs, const size_t* proof_lens) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  if (m == 0 || m > 4096 || !instances || !proofs || !proof_lens) return TG_ERR_BADARG;
+  size_t inst_stride = 32 * (size_t)c->ppk->d.n_instance_rows;
+  std::vector<PVGuard> gds(m);
+  size_t off = 0;
+  for (size_t i = 0; i < m; i++) {
+    std::vector<Fp> inst_lag;
+    int rc = pinst_from_raw(c->ppk->d, instances + inst_stride * i, inst_lag);
+    if (rc != 0) return rc;
+    rc = pverify_guard(c, *c->ppk, inst_lag, proofs + off, proof_lens[i], gds[i]);
+    if (rc != 0) return rc;
+    off += proof_lens[i];
+  }
+  std::vector<Fp> rho(m);
+  rho[0] = fd_one_mont<FpCfg>();
+  if (m > 1) {
+    Blake2b h(32, (const uint8_t*)"TaigaGPU-BatchVf");
+    uint64_t mle = (uint64_t)m;
+    h.update((const uint8_t*)&mle, 8);
+    h.update(instances, inst_stride * m);
+    h.update(proofs, off);
+    uint8_t seed[32];
+    h.final(seed);
+    Drbg rng(seed);
+    for (size_t i = 1; i < m; i++) rho[i] = rng.field<FpCfg>();
+  }
+  return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
+}
+
 /* ---- binding signatures + transaction digest (host-side wire layer,
  * SURVEY §8f-4; binding_signature.rs / transaction.rs:116-158) ----
  * RedDSA over Pallas, H* = BLAKE2b-512("Taiga_RedPallasH"), ctx-free.

@@ -90,6 +90,11 @@ def test_gpu_random_circuit_parity():
             bad = bytearray(g_proof)
             bad[-1] ^= 1
             assert not g.verify_proof_raw(inst, bytes(bad))
+            # raw-instance BATCH verify on this circuit: two valid proofs
+            # accepted together, rejected when one is tampered
+            p2 = g.create_proof_raw(inst, adv, bytes([8]) + bytes(31))
+            assert g.verify_batch_raw([(inst, g_proof), (inst, p2)])
+            assert not g.verify_batch_raw([(inst, g_proof), (inst, bytes(bad))])
         lib.orc_prover_reset()
     finally:
         g.close()
