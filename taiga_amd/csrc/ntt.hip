@@ -261,7 +261,7 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
   const Fd<FpCfg>* tw = inverse ? plan.d_tw_inv : plan.d_tw_fwd;
   u64 n = 1ULL << k;
   {
-    auto sc = prof(0);
+    [[maybe_unused]] auto sc = prof(0);
     hipLaunchKernelGGL(k_bitrev_load<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream,
                        d_tmp, d_a, k, 0, nullptr);
   }
@@ -272,7 +272,7 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
     int remaining = k - s + 1;
     if (s == 1 && remaining >= FUSE) {
       // first pass: span 1, tiles are contiguous -> 1-D variant
-      auto sc = prof(1);
+      [[maybe_unused]] auto sc = prof(1);
       u64 ntiles = n >> FUSE;
       hipLaunchKernelGGL((k_ntt_fused<FpCfg, FUSE>),
                          dim3(ntiles > 2048 ? 2048 : (unsigned)ntiles), dim3(256), 0,
@@ -283,7 +283,7 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
     int f = remaining < 7 ? remaining : 7;  // RB=8 at F=7: 33 KiB LDS
     u64 span = 1ULL << (s - 1);
     if (f >= 2 && span >= RB) {
-      auto sc = prof(1);
+      [[maybe_unused]] auto sc = prof(1);
       u64 ntiles = n >> f;
       unsigned grid = ntiles > 2048 ? 2048 : (unsigned)ntiles;
       switch (f) {
@@ -298,14 +298,14 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
       continue;
     }
     {
-      auto sc = prof(2);
+      [[maybe_unused]] auto sc = prof(2);
       hipLaunchKernelGGL(k_ntt_stage<FpCfg>, dim3(ntt_grid(n >> 1)), dim3(256), 0, stream,
                          d_tmp, tw, k, s);
     }
     s += 1;
   }
   if (inverse && ninv_mont) {
-    auto sc = prof(3);
+    [[maybe_unused]] auto sc = prof(3);
     hipLaunchKernelGGL(k_scale<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream, d_tmp, n,
                        *ninv_mont, 1, 0);
   }
