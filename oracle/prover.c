@@ -259,7 +259,7 @@ static void lag_to_coeff(fd_limbs* dst, const fd_limbs* src, int k) {
     orc_ntt_inplace((uint64_t(*)[4])dst, k, 1, FP);
 }
 
-static void coeff_to_lag(fd_limbs* dst, const fd_limbs* src, int k) {
+__attribute__((unused)) static void coeff_to_lag(fd_limbs* dst, const fd_limbs* src, int k) {
     if (dst != src) memcpy(dst, src, sizeof(fd_limbs) << k);
     orc_ntt_inplace((uint64_t(*)[4])dst, k, 0, FP);
 }
@@ -1567,7 +1567,7 @@ static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_
 
     /* multiopen protocol */
     extern int orc_multiopen_prove(const Pk* pk, tg_transcript* ts, ProofRng* rng,
-                                   const MPoly* polys, int n_polys, const MQuery* queries,
+                                   const MPoly* polys, int n_polys __attribute__((unused)), const MQuery* queries,
                                    int n_queries);
     int rc = orc_multiopen_prove(pk, &ts, &rng, polys, n_polys, queries, n_queries);
     if (rc != 0) return rc;
@@ -1604,7 +1604,7 @@ void orc_dbg_get(int which, uint8_t out[32]) {
  * DEFINED and mirrored by orc_verify. */
 
 int orc_multiopen_prove(const Pk* pk, tg_transcript* ts, ProofRng* rng,
-                        const MPoly* polys, int n_polys, const MQuery* queries,
+                        const MPoly* polys, int n_polys __attribute__((unused)), const MQuery* queries,
                         int n_queries) {
     const Desc* d = pk->d;
     long n = d->n;
