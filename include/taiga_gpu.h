@@ -180,6 +180,21 @@ int tg_tx_digest(const uint8_t* nfs, size_t n_nf, const uint8_t* cms, size_t n_c
                  const uint8_t* deltas, size_t n_delta, const uint8_t* anchors,
                  size_t n_anchor, uint8_t out[32]);
 
+/* ---- transaction wire format (layout citations in
+ *      taiga_amd/csrc/tx_wire.hpp) ----
+ * tg_tx_wire_check: ctx-free — parse a borsh Transaction, recompute
+ * Transaction::digest from the compliance instances, aggregate delta
+ * commitments into the binding vk, verify the binding signature.
+ * vk_len = byte length of one embedded RL VerifyingKey (32*(n_fixed +
+ * n_perm) for TGD1 keys). TG_OK; -1 bad binding signature; -2xx
+ * structural decode error.
+ * tg_tx_verify: the above with vk_len from the ACTIVE key, plus ONE
+ * combined batch verification of every compliance proof in the bundle
+ * (requires the active circuit's n_instance_rows <= 6). */
+int tg_tx_wire_check(const uint8_t* tx, size_t len, uint32_t vk_len,
+                     uint32_t* n_sptx, uint32_t* n_proofs);
+int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len);
+
 /* ---- kernel profiling (HIP events on the ctx stream) ----
  * names: "msm_digits", "msm_scan", "msm_scatter", "msm_bucket_acc",
  *        "msm_reduce", "msm_wsum", "ntt_stage", "ntt_fused", "ntt_bitrev",

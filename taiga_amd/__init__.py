@@ -8,11 +8,19 @@ There is NO CPU fallback here: if the extension is missing or no HIP device
 is usable, calls raise. The CPU restatement used by the tests lives in
 oracle/ and is test infrastructure only.
 """
+from . import wire  # noqa: F401
 from .api import (  # noqa: F401
+    binding_sign,
+    binding_verify,
+    binding_vk,
+    tx_digest,
+    tx_wire_check,
     TaigaGpu,
     TaigaGpuError,
     lib_path,
     load_library,
 )
 
-__all__ = ["TaigaGpu", "TaigaGpuError", "load_library", "lib_path"]
+__all__ = ["TaigaGpu", "TaigaGpuError", "load_library", "lib_path", "wire",
+           "binding_sign", "binding_verify", "binding_vk", "tx_digest",
+           "tx_wire_check"]

@@ -111,6 +111,11 @@ def load_library():
             ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
         ]
         lib.tg_tx_digest.argtypes = [ctypes.c_char_p, ctypes.c_size_t] * 4 + [ctypes.c_char_p]
+        lib.tg_tx_wire_check.argtypes = [
+            ctypes.c_char_p, ctypes.c_size_t, ctypes.c_uint32,
+            ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
+        ]
+        lib.tg_tx_verify.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
         lib.tg_prof_enable.argtypes = [ctypes.c_void_p, ctypes.c_int]
         lib.tg_prof_enable.restype = None
         lib.tg_prof_reset.argtypes = [ctypes.c_void_p]
@@ -157,6 +162,22 @@ def tx_digest(nullifiers, output_cms, delta_cms, anchors) -> bytes:
                      b"".join(delta_cms), len(delta_cms),
                      b"".join(anchors), len(anchors), out)
     return out.raw
+
+
+def tx_wire_check(tx: bytes, vk_len: int):
+    """Parse a borsh Transaction + verify its binding signature (ctx-free).
+    Returns (n_sptx, n_compliance_proofs); raises on structural errors;
+    returns None counts with valid=False on a bad signature."""
+    lib = load_library()
+    n_sptx = ctypes.c_uint32()
+    n_proofs = ctypes.c_uint32()
+    rc = lib.tg_tx_wire_check(tx, len(tx), vk_len, ctypes.byref(n_sptx),
+                              ctypes.byref(n_proofs))
+    if rc == 0:
+        return True, n_sptx.value, n_proofs.value
+    if rc == -1 or rc <= -200:
+        return False, None, None
+    raise TaigaGpuError(rc)
 
 
 def device_count():
@@ -304,6 +325,19 @@ class TaigaGpu:
         if rc == 0:
             return True
         if rc == -1 or rc <= -100 or rc == -3:
+            return False
+        raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
+
+    def tx_verify(self, tx: bytes) -> bool:
+        """Full transaction verification against the active key: wire
+        check + binding signature + one combined batch verification of all
+        compliance proofs (SURVEY §8f-4 + §8f-3 composed)."""
+        rc = self._lib.tg_tx_verify(self._h, tx, len(tx))
+        if rc == 0:
+            return True
+        if rc == -1 or rc <= -100:
+            return False
+        if rc == -3:
             return False
         raise TaigaGpuError(rc, (self._lib.tg_error_string(self._h) or b"").decode())
 

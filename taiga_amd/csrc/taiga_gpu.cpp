@@ -18,6 +18,7 @@
 #include "ntt.hip"
 #include "poseidon.hip"
 #include "binding_sig.hpp"
+#include "tx_wire.hpp"
 #include "prover_impl.hpp"
 
 namespace taiga {
@@ -815,6 +816,57 @@ int tg_tx_digest(const uint8_t* nfs, size_t n_nf, const uint8_t* cms, size_t n_c
                  size_t n_anchor, uint8_t out[32]) {
   bs_tx_digest(out, nfs, n_nf, cms, n_cm, deltas, n_delta, anchors, n_anchor);
   return TG_OK;
+}
+
+/* ---- transaction wire format (borsh layouts cited in
+ * tx_wire.hpp) ---- */
+
+/* ctx-free wire check: parse a borsh Transaction, recompute
+ * Transaction::digest from the compliance instances, aggregate delta
+ * commitments into the binding vk and verify the binding signature.
+ * vk_len = byte length of one embedded resource-logic VerifyingKey
+ * (32*(n_fixed+n_perm) for TGD1 circuits). Returns TG_OK; -1 on a bad
+ * binding signature; -2xx on structural decode errors. */
+int tg_tx_wire_check(const uint8_t* tx, size_t len, uint32_t vk_len,
+                     uint32_t* n_sptx, uint32_t* n_proofs) {
+  if (!tx || len < 4) return TG_ERR_BADARG;
+  return tx_check(tx, len, vk_len, n_sptx, n_proofs, nullptr);
+}
+
+/* full transaction verification against the ACTIVE key: wire check +
+ * binding signature + ONE combined batch verification of every compliance
+ * proof in the bundle (instances = the first n_instance_rows of each
+ * 192-byte compliance instance block; requires n_instance_rows <= 6). */
+int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  PDesc& d = c->ppk->d;
+  if (d.n_instance_rows > 6) return TG_ERR_BADARG;
+  uint32_t vk_len = 32u * (uint32_t)(d.n_fixed + d.n_perm);
+  TxDigestStreams st;
+  int rc = tx_check(tx, len, vk_len, nullptr, nullptr, &st);
+  if (rc) return rc;
+  size_t m = st.proof_ptr.size();
+  if (m == 0) return TG_OK;
+  std::vector<PVGuard> gds(m);
+  for (size_t i = 0; i < m; i++) {
+    std::vector<Fp> inst_lag;
+    rc = pinst_from_raw(d, st.inst_ptr[i], inst_lag);
+    if (rc) return rc;
+    rc = pverify_guard(c, *c->ppk, inst_lag, st.proof_ptr[i], st.proof_len[i], gds[i]);
+    if (rc) return rc;
+  }
+  std::vector<Fp> rho(m);
+  rho[0] = fd_one_mont<FpCfg>();
+  if (m > 1) {
+    Blake2b h(32, (const uint8_t*)"TaigaGPU-BatchVf");
+    h.update(tx, len);
+    uint8_t seed[32];
+    h.final(seed);
+    Drbg rng(seed);
+    for (size_t i = 1; i < m; i++) rho[i] = rng.field<FpCfg>();
+  }
+  return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
 }
 
 /* batched Poseidon P128Pow5T3 ConstantLength<L> hashing (GPU witness

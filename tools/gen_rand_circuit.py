@@ -80,7 +80,11 @@ def stack_peak(ops):
     return peak
 
 
-def gen(seed):
+def gen(seed, inst_override=None):
+    """inst_override: optional list of field values replacing the drawn
+    instance values (same rng stream is consumed either way, so the DESC
+    bytes are identical for any override — only the witness changes).
+    Extra override entries beyond n_instance_rows are ignored."""
     rng = random.Random(seed)
     bf = rng.choice([4, 5, 6])
     usable = N - (bf + 1)
@@ -98,6 +102,9 @@ def gen(seed):
     fixed = [[0] * N for _ in range(n_fixed)]
     advice = [[0] * N for _ in range(n_advice)]
     inst_vals = [rng.randrange(P) for _ in range(n_instance_rows)]
+    if inst_override is not None:
+        assert len(inst_override) >= n_instance_rows
+        inst_vals = [v % P for v in inst_override[:n_instance_rows]]
     inst_col = [0] * N
     for r, v in enumerate(inst_vals):
         inst_col[r] = v
