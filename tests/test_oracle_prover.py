@@ -48,3 +48,26 @@ def test_prove_verify_roundtrip_and_rejection():
     assert lib().orc_verify_cs1(INST, bytes(proof[:-32]), n - 32) != 0
     # wrong instance rejected
     assert lib().orc_verify_cs1(bytes([7]) + bytes(31), bytes(proof), n) != 0
+
+
+def test_mutation_rejection_sample():
+    """single-bit proof mutations at pseudo-random positions must all be
+    rejected (a full every-position sweep is run offline; this bounded
+    sample keeps the property pinned in CI)."""
+    import random
+
+    lib_ = lib()
+    inst = bytes(32)
+    wit = bytes([1]) + bytes(31)
+    rngs = bytes([2]) + bytes(31)
+    out = ctypes.create_string_buffer(1 << 14)
+    n = lib_.orc_prove_cs1(inst, wit, rngs, out, 1 << 14)
+    assert n > 0
+    proof = out.raw[:n]
+    rng = random.Random(77)
+    for _ in range(24):
+        pos = rng.randrange(n)
+        bit = 1 << rng.randrange(8)
+        bad = bytearray(proof)
+        bad[pos] ^= bit
+        assert lib_.orc_verify_cs1(inst, bytes(bad), n) != 0, f"accepted flip at {pos}"
