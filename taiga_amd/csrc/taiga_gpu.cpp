@@ -829,7 +829,7 @@ int tg_tx_digest(const uint8_t* nfs, size_t n_nf, const uint8_t* cms, size_t n_c
  * binding signature; -2xx on structural decode errors. */
 int tg_tx_wire_check(const uint8_t* tx, size_t len, uint32_t vk_len,
                      uint32_t* n_sptx, uint32_t* n_proofs) {
-  if (!tx || len < 4) return TG_ERR_BADARG;
+  if (!tx) return TG_ERR_BADARG; /* short inputs fail structurally in the parser */
   return tx_check(tx, len, vk_len, n_sptx, n_proofs, nullptr);
 }
 

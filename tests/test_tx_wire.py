@@ -165,3 +165,46 @@ def test_tx_verify_end_to_end():
         assert not g.tx_verify(bytes(bad2))
     finally:
         g.close()
+
+
+def test_wire_random_shapes_roundtrip():
+    """property: randomly shaped (but structurally valid) transactions all
+    parse with the right counts; every prefix truncation is rejected
+    structurally."""
+    import random
+
+    rng = random.Random(9)
+    for trial in range(6):
+        n_sptx = rng.randint(1, 3)
+        shapes = []
+        rs, nfs, cms, deltas, anchors = [], [], [], [], []
+        ptxs = []
+        total_proofs = 0
+        for s in range(n_sptx):
+            n_cvi = rng.randint(1, 3)
+            total_proofs += n_cvi
+            cvis = []
+            for i in range(n_cvi):
+                r = rng.randrange(1, 1 << 30)
+                rs.append(r)
+                a, nf, cm = (rng.randrange(1 << 64).to_bytes(32, "little") for _ in range(3))
+                cv = delta_commit(r)
+                cvis.append(wire.compliance_info(secrets.token_bytes(rng.randrange(0, 900)),
+                                                 wire.compliance_instance(a, nf, cm, cv)))
+                anchors.append(a), nfs.append(nf), cms.append(cm), deltas.append(cv)
+            rl = []
+            for _ in range(rng.randint(0, 2)):
+                info = wire.rl_info(secrets.token_bytes(VK_LEN), secrets.token_bytes(100),
+                                    [bytes(32)] * 22)
+                rl.append(wire.rl_set(info, [info] * rng.randint(0, 2)))
+            ptxs.append(wire.shielded_ptx(cvis, rl, rl[:1],
+                                          hints=secrets.token_bytes(rng.randrange(0, 40))))
+        digest = taiga_amd.tx_digest(nfs, cms, deltas, anchors)
+        sk = sum(rs) % Q
+        sig = taiga_amd.binding_sign(sk.to_bytes(32, "little"), digest, SEED)
+        tx = wire.transaction(ptxs, sig)
+        ok, got_sptx, got_proofs = taiga_amd.tx_wire_check(tx, VK_LEN)
+        assert ok and got_sptx == n_sptx and got_proofs == total_proofs
+        # truncations never parse as valid
+        for cut in (1, 7, len(tx) // 2, len(tx) - 1):
+            assert taiga_amd.tx_wire_check(tx[:cut], VK_LEN)[0] is False
