@@ -73,7 +73,7 @@ def main():
     ap.add_argument("--workload", choices=["proof", "msm", "ntt", "verify"], default="proof")
     ap.add_argument("--streams", type=int, default=0,
                     help="concurrent proving contexts per GPU (proof workload); "
-                         "0 = auto (8 single-rank, 2 when ranks share the host quota)")
+                         "0 = auto, scaled to this rank's CPU share")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -110,7 +110,12 @@ def main():
         golden = pathlib.Path(REPO) / "tests" / "golden"
         srs_bytes = (golden / "params_15").read_bytes()
         desc_bytes = (golden / "cs1.desc").read_bytes()
-        C = args.streams if args.streams > 0 else (8 if _local_world <= 1 else 2)
+        # auto stream count: 8 concurrent contexts when this rank has a
+        # full CPU share (single rank, or an 8-GPU node with a big cgroup
+        # quota); fall back when ranks split a small quota (host witness/
+        # transcript stages would oversubscribe and stall)
+        C = args.streams if args.streams > 0 else (
+            8 if N_CORES >= 12 else (4 if N_CORES >= 6 else 2))
         args.streams = C
         ctxs = [gpu] + [taiga_amd.TaigaGpu(local_rank) for _ in range(C - 1)]
         for g in ctxs:
