@@ -189,6 +189,15 @@ static int set_err(Ctx* c, const char* where, hipError_t e) {
   return TG_ERR_HIP;
 }
 
+// bind the calling thread to this ctx's device: HIP device selection is
+// per-thread, and callers may invoke entries from worker threads (one
+// process per GPU with thread pools — the 8-GPU bench path)
+static int tg_enter(Ctx* c) {
+  hipError_t e = hipSetDevice(c->device);
+  if (e != hipSuccess) return set_err(c, "hipSetDevice", e);
+  return 0;
+}
+
 struct ProfScope {
   Ctx* c;
   int idx;
@@ -251,6 +260,7 @@ int tg_init(int device, tg_ctx** out) {
 
 void tg_destroy(tg_ctx* ctx) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return;
   if (!c) return;
   hipStreamSynchronize(c->stream);
   for (PPk* pk : c->ppk_slots) {
@@ -274,6 +284,7 @@ void tg_destroy(tg_ctx* ctx) {
 
 int tg_synchronize(tg_ctx* ctx) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   hipError_t e = hipStreamSynchronize(c->stream);
   if (e != hipSuccess) return set_err(c, "sync", e);
   return TG_OK;
@@ -283,6 +294,7 @@ void tg_prof_enable(tg_ctx* ctx, int on) { ((Ctx*)ctx)->prof = on != 0; }
 
 void tg_prof_reset(tg_ctx* ctx) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return;
   for (auto& pc : c->prof_c) {
     for (auto ev : pc.starts) hipEventDestroy(ev);
     for (auto ev : pc.stops) hipEventDestroy(ev);
@@ -295,6 +307,7 @@ void tg_prof_reset(tg_ctx* ctx) {
 
 int tg_prof_get(tg_ctx* ctx, const char* name, double* total_ms, long* count) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   hipStreamSynchronize(c->stream);
   for (int i = 0; i < PROF_N; i++) {
     if (strcmp(PROF_NAMES[i], name) != 0) continue;
@@ -320,6 +333,7 @@ int tg_prof_get(tg_ctx* ctx, const char* name, double* total_ms, long* count) {
 
 int tg_load_srs(tg_ctx* ctx, const uint8_t* bytes, size_t len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (len < 4) return TG_ERR_BADARG;
   uint32_t k;
   memcpy(&k, bytes, 4);
@@ -374,6 +388,7 @@ int tg_srs_k(const tg_ctx* ctx) {
 
 int tg_bases_upload(tg_ctx* ctx, const uint8_t* points_xy, size_t n) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!n) return TG_ERR_BADARG;
   hipError_t e;
   if (c->d_bases && c->n_bases < n) { hipFree(c->d_bases); c->d_bases = nullptr; }
@@ -403,6 +418,7 @@ int tg_bases_upload(tg_ctx* ctx, const uint8_t* points_xy, size_t n) {
 
 int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!n) return TG_ERR_BADARG;
   hipError_t e;
   if (c->d_bases && c->n_bases < n) { hipFree(c->d_bases); c->d_bases = nullptr; }
@@ -420,6 +436,7 @@ int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed) {
 
 int tg_scalars_upload(tg_ctx* ctx, const uint8_t* scalars, size_t n) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!n) return TG_ERR_BADARG;
   hipError_t e;
   if (c->d_scalars && c->n_scalars < n) { hipFree(c->d_scalars); c->d_scalars = nullptr; }
@@ -520,6 +537,7 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
 
 int tg_msm_resident(tg_ctx* ctx, size_t n, int base_set, uint8_t out_xy[64]) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (c->n_scalars < n) return TG_ERR_STATE;
   return msm_common(c, n, base_set, out_xy);
 }
@@ -535,6 +553,7 @@ int tg_msm_pallas(tg_ctx* ctx, const uint8_t* scalars, size_t n, int base_set,
 
 int tg_poly_upload(tg_ctx* ctx, const uint8_t* poly, uint32_t k) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (k > 28) return TG_ERR_BADARG;
   u64 n = 1ULL << k;
   hipError_t e;
@@ -565,6 +584,7 @@ int tg_poly_upload(tg_ctx* ctx, const uint8_t* poly, uint32_t k) {
 
 int tg_ntt_resident(tg_ctx* ctx, int dir, uint32_t k, int coset) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (coset) return TG_ERR_BADARG;  // arrives with the prover pipeline
   if (c->poly_k != (int)k) return TG_ERR_STATE;
   hipError_t e;
@@ -592,6 +612,7 @@ int tg_ntt_resident(tg_ctx* ctx, int dir, uint32_t k, int coset) {
 
 int tg_poly_download(tg_ctx* ctx, uint8_t* poly, uint32_t k) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (c->poly_k != (int)k) return TG_ERR_STATE;
   u64 n = 1ULL << k;
   // from_mont into tmp, then D2H
@@ -622,6 +643,7 @@ extern "C" {
  * proof (resource_logic_circuit.rs:578-580); the cache is §8f-1's fix. */
 int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (c->k < 0) return TG_ERR_NOSRS;
   PPk* pk = new PPk();
   int rc = ppk_keygen(c, *pk, desc, desc_len);
@@ -637,6 +659,7 @@ int tg_keygen(tg_ctx* ctx, const uint8_t* desc, size_t desc_len) {
 /* select a previously built proving key by slot id */
 int tg_select_key(tg_ctx* ctx, int slot) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (slot < 0 || (size_t)slot >= c->ppk_slots.size()) return TG_ERR_BADARG;
   c->ppk = c->ppk_slots[slot];
   return TG_OK;
@@ -646,6 +669,7 @@ int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_
                     const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
                     size_t* out_len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   std::vector<uint8_t> proof;
   int rc = pprove(c, *c->ppk, inst_seed, wit_seed, rng_seed, proof);
@@ -659,6 +683,7 @@ int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_
 int tg_witness_hash(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t wit_seed[32],
                     uint8_t out[32]) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   PDesc& d = c->ppk->d;
   std::vector<Fp> inst;
@@ -688,6 +713,7 @@ extern "C" {
 int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* proof,
                     size_t proof_len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   return pverify(c, *c->ppk, inst_seed, proof, proof_len);
 }
@@ -698,6 +724,7 @@ int tg_verify_proof(tg_ctx* ctx, const uint8_t inst_seed[32], const uint8_t* pro
 int tg_verify_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* proof,
                         size_t proof_len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   if (!instance) return TG_ERR_BADARG;
   return pverify_raw(c, *c->ppk, instance, proof, proof_len);
@@ -714,6 +741,7 @@ int tg_verify_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* pro
 int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
                     const uint8_t* proofs, const size_t* proof_lens) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   if (m == 0 || m > 4096 || !inst_seeds || !proofs || !proof_lens) return TG_ERR_BADARG;
   std::vector<PVGuard> gds(m);
@@ -750,6 +778,7 @@ int tg_verify_batch(tg_ctx* ctx, size_t m, const uint8_t* inst_seeds,
 int tg_verify_batch_raw(tg_ctx* ctx, size_t m, const uint8_t* instances,
                         const uint8_t* proofs, const size_t* proof_lens) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   if (m == 0 || m > 4096 || !instances || !proofs || !proof_lens) return TG_ERR_BADARG;
   size_t inst_stride = 32 * (size_t)c->ppk->d.n_instance_rows;
@@ -839,6 +868,7 @@ int tg_tx_wire_check(const uint8_t* tx, size_t len, uint32_t vk_len,
  * 192-byte compliance instance block; requires n_instance_rows <= 6). */
 int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   PDesc& d = c->ppk->d;
   if (d.n_instance_rows > 6) return TG_ERR_BADARG;
@@ -876,6 +906,7 @@ int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len) {
 int tg_poseidon_hash(tg_ctx* ctx, const uint8_t* msgs, size_t n, int L,
                      uint8_t* out) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!msgs || !out || n == 0 || L < 1 || L > 64) return TG_ERR_BADARG;
   hipError_t e = hipSetDevice(c->device);
   if (e != hipSuccess) return set_err(c, "hipSetDevice", e);
@@ -923,6 +954,7 @@ int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* adv
                         const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
                         size_t* out_len) {
   Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
   if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
   std::vector<uint8_t> proof;
   int rc = pprove_raw(c, *c->ppk, instance, advice, rng_seed, proof);
