@@ -19,6 +19,7 @@
 #include "poseidon.hip"
 #include "binding_sig.hpp"
 #include "tx_wire.hpp"
+#include "fd28.hpp"
 #include "prover_impl.hpp"
 
 namespace taiga {
@@ -806,6 +807,30 @@ int tg_verify_batch_raw(tg_ctx* ctx, size_t m, const uint8_t* instances,
     for (size_t i = 1; i < m; i++) rho[i] = rng.field<FpCfg>();
   }
   return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
+}
+
+/* experiment self-test (fd28.hpp — carry-chain-free radix-2^28
+ * Montgomery multiply, round-2 kernel candidate): host-side evaluation of
+ * out = a * b * 2^-280 mod p over canonical Fp bytes. The arithmetic is
+ * host/device-shared TG_HD code, so this pins the device semantics
+ * numerically without a GPU (tests/test_fd28.py). Not a product entry. */
+int tg_dbg_fd28_mul(const uint8_t a_bytes[32], const uint8_t b_bytes[32],
+                    uint8_t out[32]) {
+  Fp a, b;
+  memcpy(a.l, a_bytes, 32);
+  memcpy(b.l, b_bytes, 32);
+  auto canon = [](const Fp& v) {
+    for (int limb = 3; limb >= 0; limb--) {
+      if (v.l[limb] > FpCfg::MOD[limb]) return false;
+      if (v.l[limb] < FpCfg::MOD[limb]) return true;
+    }
+    return false;  // equal to p
+  };
+  if (!canon(a) || !canon(b)) return TG_ERR_ENCODING;
+  Fd28<FpCfg> r = fd28_mul(fd28_from<FpCfg>(a), fd28_from<FpCfg>(b));
+  Fp v = fd28_norm(r);
+  memcpy(out, v.l, 32);
+  return TG_OK;
 }
 
 /* ---- binding signatures + transaction digest (host-side wire layer,

@@ -90,3 +90,23 @@ __global__ void k3(Fd* out, const Fd* in, int n) {
   Fd r1 = mul_addc(c, d);
   out[2*i] = r0; out[2*i+1] = r1;
 }
+
+// k4: the fd28 carry-chain-free Montgomery multiply (taiga_amd/csrc/
+// fd28.hpp) on the same workload shape as k1 — objdump slot comparison.
+#include "../../taiga_amd/csrc/fd28.hpp"
+using namespace taiga;
+__global__ void k4(Fd28<FpCfg>* out, const Fd28<FpCfg>* in, int n) {
+  int i = blockIdx.x * 256 + threadIdx.x;
+  Fd28<FpCfg> a = in[2 * i], b = in[2 * i + 1];
+  out[i] = fd28_mul(a, b);
+}
+
+// k5: two independent fd28 muls (interleaving headroom without VCC)
+__global__ void k5(Fd28<FpCfg>* out, const Fd28<FpCfg>* in, int n) {
+  int i = blockIdx.x * 256 + threadIdx.x;
+  Fd28<FpCfg> a = in[4 * i], b = in[4 * i + 1], c = in[4 * i + 2], d = in[4 * i + 3];
+  Fd28<FpCfg> r0 = fd28_mul(a, b);
+  Fd28<FpCfg> r1 = fd28_mul(c, d);
+  out[2 * i] = r0;
+  out[2 * i + 1] = r1;
+}
