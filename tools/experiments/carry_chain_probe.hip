@@ -1,3 +1,12 @@
+// Compile-only probes for the Montgomery-multiply instruction stream
+// (see profiles/r01_bucket_acc_hazard_analysis.txt for the findings):
+//   hipcc --offload-arch=gfx950 -O3 -I../../taiga_amd/csrc -c carry_chain_probe.hip
+// then llvm-objdump the gfx950 bundle and count s_nop wait states.
+// k1/k2: current u128 CIOS (one / two independent muls)
+// k3: __builtin_addcll lowering (dead end: more real instructions)
+// k4/k5: fd28 lazy-carry radix-2^28 (32% fewer slots; measured EQUAL wall
+//        time at full occupancy -> multiply is real-op bound, not hazard
+//        bound; see fd28_bench.hip for the dynamic A/B)
 #include <hip/hip_runtime.h>
 typedef unsigned long long u64;
 typedef unsigned __int128 u128;
