@@ -889,8 +889,12 @@ int tg_tx_wire_check(const uint8_t* tx, size_t len, uint32_t vk_len,
 
 /* full transaction verification against the ACTIVE key: wire check +
  * binding signature + ONE combined batch verification of every compliance
- * proof in the bundle (instances = the first n_instance_rows of each
- * 192-byte compliance instance block; requires n_instance_rows <= 6). */
+ * proof in the bundle. Round-1 instance mapping: the first
+ * n_instance_rows fields of each 192-byte compliance instance block
+ * (requires n_instance_rows <= 6). The REAL compliance circuit expands
+ * the 6 borsh fields into 9 instance rows — [nf, anchor, cm, delta.x,
+ * delta.y, rl_in x2 halves, rl_out x2 halves], constant.rs:54-62 — which
+ * replaces this mapping in round 2. */
 int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len) {
   Ctx* c = (Ctx*)ctx;
   if (!c || tg_enter(c)) return TG_ERR_BADARG;
