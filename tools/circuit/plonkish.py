@@ -1,0 +1,817 @@
+"""PLONKish constraint-system model + region layouter + witness tracing.
+
+Round-2 restatement toolchain (see fields.py header). This module mirrors,
+from scratch, the halo2_proofs 0.3 circuit-construction machinery the
+un-vendored heliaxdev/halo2 dep provides to the reference circuits
+(SURVEY.md §8c):
+
+  - ConstraintSystem: advice/fixed/instance columns, simple & complex
+    selectors, gates, one-expression lookups, copy constraints
+    (permutation cycles, spliced exactly like halo2
+    permutation::Assembly::copy), enable_constant.
+  - Region/Layouter: regions with relative offsets packed by a
+    per-column-watermark first-fit (column-max) packer. The reference uses
+    floor_planner::V1 (compliance_circuit.rs:71); bit-identical row layout
+    vs the Rust binary is unpinnable in-container (SURVEY §8c residual
+    risk), so the layout here is self-consistent and the CS-level counts
+    (columns, queries, degree) are what parity rests on.
+  - Selector compression: restated from halo2 0.3
+    plonk/circuit/compress_selectors.rs `process` (exclusion matrix over
+    activation overlap; greedy in-order packing bounded by
+    content-degree + combination length <= cs degree; per-member indicator
+    expression q * prod_{j!=i}(code_j - q) normalized to 1 at code_i).
+  - Emission: TGD2 circuit-description blob (TGD1 + per-constraint gating
+    hints + larger capacities) consumed by BOTH provers, and a TGW1
+    straight-line witness-synthesis program interpreted at prove time.
+  - MockProver-equivalent verification (gates vanish, lookups contained,
+    copies consistent, instance exposure).
+
+Witness tracing: all witness values are V objects (SSA registers over Fp).
+Circuit synthesize() code must be branch-free on witness values (use
+select()/is_zero()) so the emitted program is input-independent.
+"""
+from __future__ import annotations
+
+import struct
+from dataclasses import dataclass, field
+
+from . import fields as F
+
+# ---------------------------------------------------------------- expressions
+
+
+class Expr:
+    def __add__(self, o):
+        return Sum(self, _lift(o))
+
+    def __radd__(self, o):
+        return Sum(_lift(o), self)
+
+    def __sub__(self, o):
+        return Sub(self, _lift(o))
+
+    def __rsub__(self, o):
+        return Sub(_lift(o), self)
+
+    def __mul__(self, o):
+        o = _lift(o)
+        if isinstance(o, Const):
+            return Scaled(self, o.v)
+        return Prod(self, o)
+
+    def __rmul__(self, o):
+        return _lift(o) * self
+
+    def __neg__(self):
+        return Neg(self)
+
+    def square(self):
+        return self * self
+
+
+def _lift(x):
+    if isinstance(x, Expr):
+        return x
+    return Const(x % F.P)
+
+
+@dataclass(frozen=True)
+class Const(Expr):
+    v: int
+
+
+@dataclass(frozen=True)
+class FixedQ(Expr):
+    col: "Column"
+    rot: int
+
+
+@dataclass(frozen=True)
+class AdviceQ(Expr):
+    col: "Column"
+    rot: int
+
+
+@dataclass(frozen=True)
+class InstanceQ(Expr):
+    col: "Column"
+    rot: int
+
+
+@dataclass(frozen=True)
+class SelQ(Expr):
+    sel: "Selector"
+
+
+@dataclass(frozen=True)
+class Sum(Expr):
+    a: Expr
+    b: Expr
+
+
+@dataclass(frozen=True)
+class Sub(Expr):
+    a: Expr
+    b: Expr
+
+
+@dataclass(frozen=True)
+class Prod(Expr):
+    a: Expr
+    b: Expr
+
+
+@dataclass(frozen=True)
+class Neg(Expr):
+    a: Expr
+
+
+@dataclass(frozen=True)
+class Scaled(Expr):
+    a: Expr
+    v: int
+
+
+def expr_degree(e: Expr) -> int:
+    if isinstance(e, Const):
+        return 0
+    if isinstance(e, (FixedQ, AdviceQ, InstanceQ, SelQ)):
+        return 1
+    if isinstance(e, (Sum, Sub)):
+        return max(expr_degree(e.a), expr_degree(e.b))
+    if isinstance(e, Prod):
+        return expr_degree(e.a) + expr_degree(e.b)
+    if isinstance(e, (Neg, Scaled)):
+        return expr_degree(e.a)
+    raise TypeError(e)
+
+
+def subst_selectors(e: Expr, mapping) -> Expr:
+    if isinstance(e, SelQ):
+        return mapping[e.sel]
+    if isinstance(e, Sum):
+        return Sum(subst_selectors(e.a, mapping), subst_selectors(e.b, mapping))
+    if isinstance(e, Sub):
+        return Sub(subst_selectors(e.a, mapping), subst_selectors(e.b, mapping))
+    if isinstance(e, Prod):
+        return Prod(subst_selectors(e.a, mapping), subst_selectors(e.b, mapping))
+    if isinstance(e, Neg):
+        return Neg(subst_selectors(e.a, mapping))
+    if isinstance(e, Scaled):
+        return Scaled(subst_selectors(e.a, mapping), e.v)
+    return e
+
+
+def collect_queries(e: Expr, out):
+    if isinstance(e, FixedQ):
+        out["fixed"].add((e.col, e.rot))
+    elif isinstance(e, AdviceQ):
+        out["advice"].add((e.col, e.rot))
+    elif isinstance(e, InstanceQ):
+        out["instance"].add((e.col, e.rot))
+    elif isinstance(e, (Sum, Sub, Prod)):
+        collect_queries(e.a, out)
+        collect_queries(e.b, out)
+    elif isinstance(e, (Neg, Scaled)):
+        collect_queries(e.a, out)
+    elif isinstance(e, SelQ):
+        raise ValueError("selector not substituted")
+
+
+# postfix emission -----------------------------------------------------------
+
+OP_CONST, OP_FIXED, OP_ADVICE, OP_INSTANCE, OP_ADD, OP_SUB, OP_MUL, OP_NEG, OP_SCALE = range(9)
+
+
+class PostfixEmitter:
+    """Emits TGD postfix ops, minimizing stack depth (deeper operand first,
+    using commutativity / a-b = -(b) + a rewrites)."""
+
+    def __init__(self, const_idx, fixed_idx, advice_idx, instance_idx):
+        self.ci = const_idx
+        self.fi = fixed_idx
+        self.ai = advice_idx
+        self.ii = instance_idx
+        self.ops = []
+        self.max_depth = 0
+
+    def _leaf_depth(self, e):
+        return 1
+
+    def depth(self, e):
+        if isinstance(e, (Const, FixedQ, AdviceQ, InstanceQ)):
+            return 1
+        if isinstance(e, (Neg, Scaled)):
+            return self.depth(e.a)
+        a, b = self.depth(e.a), self.depth(e.b)
+        lo, hi = min(a, b), max(a, b)
+        return max(hi, lo + 1)
+
+    def emit(self, e, d=0):
+        if isinstance(e, Const):
+            self.ops.append((OP_CONST, self.ci(e.v), 0))
+            self._bump(d + 1)
+        elif isinstance(e, FixedQ):
+            self.ops.append((OP_FIXED, self.fi(e.col), e.rot))
+            self._bump(d + 1)
+        elif isinstance(e, AdviceQ):
+            self.ops.append((OP_ADVICE, self.ai(e.col), e.rot))
+            self._bump(d + 1)
+        elif isinstance(e, InstanceQ):
+            self.ops.append((OP_INSTANCE, self.ii(e.col), e.rot))
+            self._bump(d + 1)
+        elif isinstance(e, Neg):
+            self.emit(e.a, d)
+            self.ops.append((OP_NEG, 0, 0))
+        elif isinstance(e, Scaled):
+            self.emit(e.a, d)
+            self.ops.append((OP_SCALE, self.ci(e.v), 0))
+        elif isinstance(e, Sum):
+            x, y = e.a, e.b
+            if self.depth(y) > self.depth(x):
+                x, y = y, x
+            self.emit(x, d)
+            self.emit(y, d + 1)
+            self.ops.append((OP_ADD, 0, 0))
+        elif isinstance(e, Prod):
+            x, y = e.a, e.b
+            if self.depth(y) > self.depth(x):
+                x, y = y, x
+            self.emit(x, d)
+            self.emit(y, d + 1)
+            self.ops.append((OP_MUL, 0, 0))
+        elif isinstance(e, Sub):
+            if self.depth(e.b) > self.depth(e.a):
+                self.emit(e.b, d)
+                self.ops.append((OP_NEG, 0, 0))
+                self.emit(e.a, d + 1)
+                self.ops.append((OP_ADD, 0, 0))
+            else:
+                self.emit(e.a, d)
+                self.emit(e.b, d + 1)
+                self.ops.append((OP_SUB, 0, 0))
+        else:
+            raise TypeError(e)
+
+    def _bump(self, d):
+        if d > self.max_depth:
+            self.max_depth = d
+
+
+# ---------------------------------------------------------------- witness SSA
+
+# TGW1 opcodes
+W_LOADI, W_CONST, W_ADD, W_SUB, W_MUL, W_INV0, W_NEG, W_SQRT0, W_BIT, W_BYTE = range(10)
+
+
+class Prog:
+    """SSA witness-synthesis program with eager concrete evaluation."""
+
+    def __init__(self, n_inputs):
+        self.n_inputs = n_inputs
+        self.ops = []  # (opcode, a, b) -> one reg each, reg id = index
+        self.vals = []  # concrete value per reg (for the sample input)
+        self.consts = []  # const table
+        self._cmap = {}
+        self._memo = {}
+        self.input_vals = None  # set by run()
+
+    def set_inputs(self, vals):
+        assert len(vals) == self.n_inputs
+        self.input_vals = [v % F.P for v in vals]
+        # re-evaluate whole program (used when re-running with new inputs)
+        self.vals = []
+        for (op, a, b) in self.ops:
+            self.vals.append(self._eval(op, a, b))
+
+    def _eval(self, op, a, b):
+        v = self.vals
+        if op == W_LOADI:
+            return self.input_vals[a]
+        if op == W_CONST:
+            return self.consts[a]
+        if op == W_ADD:
+            return (v[a] + v[b]) % F.P
+        if op == W_SUB:
+            return (v[a] - v[b]) % F.P
+        if op == W_MUL:
+            return (v[a] * v[b]) % F.P
+        if op == W_INV0:
+            return F.inv0(v[a])
+        if op == W_NEG:
+            return (-v[a]) % F.P
+        if op == W_SQRT0:
+            return F.sqrt0(v[a])
+        if op == W_BIT:
+            return F.fbit(v[a], b)
+        if op == W_BYTE:
+            return F.fbyte(v[a], b)
+        raise ValueError(op)
+
+    def _push(self, op, a, b):
+        key = (op, a, b)
+        r = self._memo.get(key)
+        if r is not None:
+            return r
+        self.ops.append(key)
+        self.vals.append(self._eval(op, a, b))
+        r = len(self.ops) - 1
+        self._memo[key] = r
+        return r
+
+    def load_input(self, i):
+        return V(self, self._push(W_LOADI, i, 0))
+
+    def const(self, c):
+        c %= F.P
+        ci = self._cmap.get(c)
+        if ci is None:
+            ci = len(self.consts)
+            self.consts.append(c)
+            self._cmap[c] = ci
+        return V(self, self._push(W_CONST, ci, 0))
+
+
+class V:
+    """Traced witness value (SSA register)."""
+
+    __slots__ = ("p", "r")
+
+    def __init__(self, p, r):
+        self.p = p
+        self.r = r
+
+    @property
+    def v(self):
+        return self.p.vals[self.r]
+
+    def _c(self, o):
+        if isinstance(o, V):
+            return o
+        return self.p.const(o)
+
+    def __add__(self, o):
+        o = self._c(o)
+        return V(self.p, self.p._push(W_ADD, self.r, o.r))
+
+    __radd__ = __add__
+
+    def __sub__(self, o):
+        o = self._c(o)
+        return V(self.p, self.p._push(W_SUB, self.r, o.r))
+
+    def __rsub__(self, o):
+        return self._c(o) - self
+
+    def __mul__(self, o):
+        o = self._c(o)
+        return V(self.p, self.p._push(W_MUL, self.r, o.r))
+
+    __rmul__ = __mul__
+
+    def __neg__(self):
+        return V(self.p, self.p._push(W_NEG, self.r, 0))
+
+    def inv0(self):
+        return V(self.p, self.p._push(W_INV0, self.r, 0))
+
+    def sqrt0(self):
+        return V(self.p, self.p._push(W_SQRT0, self.r, 0))
+
+    def bit(self, i):
+        return V(self.p, self.p._push(W_BIT, self.r, i))
+
+    def byte(self, i):
+        return V(self.p, self.p._push(W_BYTE, self.r, i))
+
+    def square(self):
+        return self * self
+
+    def is_zero(self):
+        # 1 - x * inv0(x): 1 iff x == 0
+        return 1 - self * self.inv0()
+
+    def select(self, a, b):
+        """self is 0/1: self ? a : b  (arithmetic, branch-free)."""
+        a = self._c(a)
+        b = self._c(b)
+        return b + self * (a - b)
+
+
+# ---------------------------------------------------------------- CS objects
+
+
+@dataclass(frozen=True)
+class Column:
+    kind: str  # "advice" | "fixed" | "instance" | "table"
+    index: int
+
+    def cur(self):
+        return self.q(0)
+
+    def next(self):
+        return self.q(1)
+
+    def prev(self):
+        return self.q(-1)
+
+    def q(self, rot):
+        if self.kind == "advice":
+            return AdviceQ(self, rot)
+        if self.kind in ("fixed", "table"):
+            return FixedQ(self, rot)
+        return InstanceQ(self, rot)
+
+
+class Selector:
+    _n = 0
+
+    def __init__(self, cs, simple):
+        self.cs = cs
+        self.simple = simple
+        self.id = Selector._n
+        Selector._n += 1
+        self.rows = set()
+        self.max_content_degree = 0  # max gate-content degree (sans selector)
+
+    def expr(self):
+        return SelQ(self)
+
+    def enable(self, region, offset):
+        region._sel.append((self, offset))
+
+
+@dataclass
+class Gate:
+    name: str
+    cname: str
+    sel: Selector  # gating selector (every reference gate is sel * content)
+    content: Expr
+
+
+class Cell:
+    __slots__ = ("col", "row", "reg")
+
+    def __init__(self, col, row, reg):
+        self.col = col
+        self.row = row
+        self.reg = reg
+
+    def value(self):
+        return self.reg
+
+    @property
+    def v(self):
+        return self.reg.v
+
+
+class Region:
+    def __init__(self, cs, name):
+        self.cs = cs
+        self.name = name
+        self._adv = []  # (col, offset, V, is_copy_of Cell|None, const|None)
+        self._fix = []  # (col, offset, int)
+        self._sel = []  # (Selector, offset)
+        self.cells = []
+
+    def assign_advice(self, col, offset, val: V) -> Cell:
+        c = Cell(col, None, val)
+        self._adv.append((col, offset, c, None, None))
+        return c
+
+    def copy_advice(self, cell: Cell, col, offset) -> Cell:
+        c = Cell(col, None, cell.reg)
+        self._adv.append((col, offset, c, cell, None))
+        return c
+
+    def assign_advice_from_constant(self, col, offset, cval: int) -> Cell:
+        cval %= F.P
+        c = Cell(col, None, self.cs.prog.const(cval))
+        self._adv.append((col, offset, c, None, cval))
+        return c
+
+    def assign_fixed(self, col, offset, cval: int):
+        self._fix.append((col, offset, cval % F.P))
+
+    def constrain_equal(self, a: Cell, b: Cell):
+        # both cells already absolutized (from earlier regions) or in this one:
+        # defer to cs at close-time
+        self.cs._pending_eq.append((a, b))
+
+    def assign_advice_from_instance(self, icol, irow, col, offset) -> Cell:
+        v = self.cs.instance_v[irow]
+        c = Cell(col, None, v)
+        self._adv.append((col, offset, c, ("inst", irow), None))
+        return c
+
+
+class ConstraintSystem:
+    def __init__(self, k, name="circuit"):
+        self.k = k
+        self.n = 1 << k
+        self.name = name
+        self.advice_cols = []
+        self.fixed_cols = []  # user fixed columns (incl table cols)
+        self.instance_cols = []
+        self.selectors = []
+        self.gates = []  # list[Gate]
+        self.lookups = []  # (name, [input exprs], [table exprs])
+        self.eq_cols = []  # permutation columns in enable order
+        self._eq_set = set()
+        self.const_col = None
+        self._const_cursor = 0
+        # layout
+        self._watermark = {}  # column/"sel:<id>" -> next free row
+        self.advice_vals = None  # filled at synth finalize: [col][row] -> reg or None
+        self.fixed_vals = None  # [col][row] int
+        self.copies = []  # ((colkindidx, row), (colkindidx, row))
+        self.prog = None
+        self.instance_v = None
+        self.regions = 0
+        self._pending_eq = []
+        self.table_rows = {}  # table col -> n assigned rows
+
+    # -- configure-phase API
+    def advice_column(self):
+        c = Column("advice", len(self.advice_cols))
+        self.advice_cols.append(c)
+        return c
+
+    def fixed_column(self):
+        c = Column("fixed", len(self.fixed_cols))
+        self.fixed_cols.append(c)
+        return c
+
+    def lookup_table_column(self):
+        c = Column("table", len(self.fixed_cols))
+        self.fixed_cols.append(c)
+        return c
+
+    def instance_column(self):
+        c = Column("instance", len(self.instance_cols))
+        self.instance_cols.append(c)
+        return c
+
+    def selector(self):
+        s = Selector(self, True)
+        self.selectors.append(s)
+        return s
+
+    def complex_selector(self):
+        s = Selector(self, False)
+        self.selectors.append(s)
+        return s
+
+    def enable_equality(self, col):
+        if col not in self._eq_set:
+            self._eq_set.add(col)
+            self.eq_cols.append(col)
+
+    def enable_constant(self, col):
+        self.enable_equality(col)
+        self.const_col = col
+
+    def create_gate(self, name, sel, constraints):
+        """constraints: list of (cname, content_expr); gate poly = sel * content."""
+        for cname, content in constraints:
+            d = expr_degree(content)
+            sel.max_content_degree = max(sel.max_content_degree, d)
+            self.gates.append(Gate(name, cname, sel, content))
+
+    def add_lookup(self, name, inputs, tables):
+        self.lookups.append((name, inputs, tables))
+
+    # -- synthesize-phase API
+    def start_synth(self, prog: Prog, instance_vals):
+        self.prog = prog
+        self.instance_v = instance_vals  # list of V (loaded from inputs)
+        self.advice_vals = [dict() for _ in self.advice_cols]
+        self.fixed_vals = [dict() for _ in self.fixed_cols]
+
+    def region(self, name):
+        return _RegionCtx(self, name)
+
+    def assign_table(self, col, values):
+        """Table column assignment: rows 0..len(values); remaining usable rows
+        filled with values[0] (halo2 DefaultTableValue = first assigned)."""
+        ci = col.index
+        for i, v in enumerate(values):
+            self.fixed_vals[ci][i] = v % F.P
+        self.table_rows[ci] = len(values)
+
+    def constrain_instance(self, cell: Cell, icol, irow):
+        self.copies.append((("instance", icol.index, irow), (cell.col, cell.row)))
+        # value check happens in mock verify
+
+    def _close_region(self, r: Region):
+        self.regions += 1
+        # footprint
+        cols = set()
+        height = 0
+        for col, off, _, _, _ in r._adv:
+            cols.add(col)
+            height = max(height, off + 1)
+        for col, off, _ in r._fix:
+            cols.add(col)
+            height = max(height, off + 1)
+        for s, off in r._sel:
+            cols.add(("sel", s.id))
+            height = max(height, off + 1)
+        start = 0
+        for c in cols:
+            start = max(start, self._watermark.get(c, 0))
+        for c in cols:
+            self._watermark[c] = start + height
+        # materialize
+        for col, off, cell, src, cval in r._adv:
+            row = start + off
+            cell.row = row
+            prev = self.advice_vals[col.index].get(row)
+            if prev is not None:
+                raise ValueError(f"advice overlap {col} row {row} in {r.name}")
+            self.advice_vals[col.index][row] = cell.reg
+            if isinstance(src, Cell):
+                self.copies.append(((src.col, src.row), (col, row)))
+            elif isinstance(src, tuple) and src and src[0] == "inst":
+                self.copies.append(
+                    (("instance", 0, src[1]), (col, row))
+                )
+            if cval is not None:
+                # halo2 assigns the constant into the constants fixed column
+                # at a fresh row and copy-constrains it
+                crow = self._alloc_const_row()
+                self.fixed_vals[self.const_col.index][crow] = cval
+                self.copies.append(((self.const_col, crow), (col, row)))
+        for col, off, cval in r._fix:
+            row = start + off
+            prev = self.fixed_vals[col.index].get(row)
+            if prev is not None and prev != cval:
+                raise ValueError(f"fixed overlap {col} row {row} in {r.name}")
+            self.fixed_vals[col.index][row] = cval
+        for s, off in r._sel:
+            s.rows.add(start + off)
+        for a, b in self._pending_eq:
+            self.copies.append(((a.col, a.row), (b.col, b.row)))
+        self._pending_eq = []
+
+    def _alloc_const_row(self):
+        key = self.const_col
+        row = self._watermark.get(key, 0)
+        self._watermark[key] = row + 1
+        return row
+
+    # ------------------------------------------------------------- finalize
+    def blinding_factors(self):
+        # halo2 0.3: max advice queries per column (min 3) + 1 (multiopen) + 1
+        per_col = {}
+        q = self._all_queries()
+        for (col, rot) in q["advice"]:
+            per_col.setdefault(col, set()).add(rot)
+        factors = max((len(s) for s in per_col.values()), default=1)
+        factors = max(3, factors)
+        return factors + 1 + 1
+
+    def _all_queries(self):
+        out = {"advice": set(), "fixed": set(), "instance": set()}
+        mapping = self._selector_exprs()
+        for g in self.gates:
+            collect_queries(subst_selectors(Prod(g.sel.expr(), g.content), mapping), out)
+        for name, ins, tabs in self.lookups:
+            for e in ins + tabs:
+                collect_queries(subst_selectors(e, mapping), out)
+        # permutation columns are queried at cur (halo2 adds these)
+        for col in self.eq_cols:
+            out[col.kind if col.kind != "table" else "fixed"].add((col, 0))
+        return out
+
+    def degree(self):
+        """cs.degree() AFTER selector compression (what sizes the quotient)."""
+        mapping = self._selector_exprs()
+        d = 3  # permutation / vanishing minimum
+        for g in self.gates:
+            d = max(d, expr_degree(subst_selectors(Prod(g.sel.expr(), g.content), mapping)))
+        for name, ins, tabs in self.lookups:
+            ideg = max(expr_degree(subst_selectors(e, mapping)) for e in ins)
+            tdeg = max(expr_degree(subst_selectors(e, mapping)) for e in tabs)
+            d = max(d, 2 + max(ideg, tdeg))
+        # permutation: chunk+2 where chunk = degree-2 — consistent by constr.
+        return d
+
+    def raw_degree(self):
+        """degree with selectors as degree-1 (pre-compression bound passed to
+        compress_selectors, like halo2 keygen does)."""
+        d = 3
+        for g in self.gates:
+            d = max(d, 1 + expr_degree(g.content))
+        for name, ins, tabs in self.lookups:
+            # selectors in lookup exprs are complex -> degree 1 columns
+            mapping = {s: FixedQ(Column("fixed", 10_000 + s.id), 0) for s in self.selectors}
+            ideg = max(expr_degree(subst_selectors(e, mapping)) for e in ins)
+            tdeg = max(expr_degree(subst_selectors(e, mapping)) for e in tabs)
+            d = max(d, 2 + max(ideg, tdeg))
+        return d
+
+    # selector compression (halo2 0.3 compress_selectors::process) ----------
+    def compress_selectors(self):
+        """Returns nothing; caches mapping sel -> Expr over new fixed columns
+        and appends combination columns to self.fixed_cols with values."""
+        if getattr(self, "_sel_map", None) is not None:
+            return
+        max_degree = self.raw_degree()
+        n = self.n
+        sel_map = {}
+        combo_cols = []  # (Column, values dict)
+
+        descs = list(self.selectors)  # creation order
+        # degree-0 (complex or unused) selectors first, own columns
+        simple = []
+        for s in descs:
+            if (not s.simple) or s.max_content_degree == 0:
+                col = self.fixed_column()
+                vals = {r: 1 for r in s.rows}
+                self.fixed_vals.append(dict())
+                self.fixed_vals[col.index] = vals
+                sel_map[s] = FixedQ(col, 0)
+                combo_cols.append(col)
+            else:
+                simple.append(s)
+
+        added = [False] * len(simple)
+        for i, s in enumerate(simple):
+            if added[i]:
+                continue
+            added[i] = True
+            d = s.max_content_degree  # content degree (selector omitted)
+            combination = [s]
+            idxs = [i]
+            for j in range(i + 1, len(simple)):
+                if d + len(combination) >= max_degree:
+                    break
+                if added[j]:
+                    continue
+                t = simple[j]
+                # exclusion: overlapping activations cannot combine
+                if any(not t.rows.isdisjoint(simple[x].rows) for x in idxs):
+                    continue
+                t_d = t.max_content_degree
+                if max(d, t_d) + len(combination) + 1 > max_degree:
+                    continue
+                d = max(d, t_d)
+                combination.append(t)
+                idxs.append(j)
+                added[j] = True
+            col = self.fixed_column()
+            self.fixed_vals.append(dict())
+            vals = {}
+            m = len(combination)
+            q = FixedQ(col, 0)
+            for pos, t in enumerate(combination):
+                code = pos + 1
+                for r in t.rows:
+                    vals[r] = code
+                if m == 1:
+                    sel_map[t] = q
+                else:
+                    e: Expr = q
+                    norm = code
+                    for other in range(1, m + 1):
+                        if other == code:
+                            continue
+                        e = Prod(e, Sub(Const(other), q))
+                        norm = norm * (other - code) % F.P
+                    sel_map[t] = Scaled(e, F.inv0(norm))
+            self.fixed_vals[col.index] = vals
+            combo_cols.append(col)
+        self._sel_map = sel_map
+
+    def _selector_exprs(self):
+        self.compress_selectors()
+        return self._sel_map
+
+
+class _RegionCtx:
+    def __init__(self, cs, name):
+        self.cs = cs
+        self.r = Region(cs, name)
+
+    def __enter__(self):
+        return self.r
+
+    def __exit__(self, et, ev, tb):
+        if et is None:
+            self.cs._close_region(self.r)
+        return False
+
+
+# simple namespace helpers mirroring common halo2 call shapes ---------------
+
+
+def assign_free_advice(cs, col, val: V) -> Cell:
+    """gadgets.rs assign_free_advice: one-row region."""
+    with cs.region("load private") as r:
+        return r.assign_advice(col, 0, val)
+
+
+def assign_free_constant(cs, col, cval: int) -> Cell:
+    with cs.region("load constant") as r:
+        return r.assign_advice_from_constant(col, 0, cval)
