@@ -36,13 +36,17 @@ void orc_drbg_raw(const uint8_t seed[32], long n, uint8_t* out);
 #define FQ (&FD_Q)
 
 static void pallas_gen(pt_jac* g) {
-    /* (-1, 2) in Mont form over Fp */
+    /* RESOURCE_COMMIT_DOMAIN.R() (constant.rs:160): sinsemilla
+     * CommitDomain("Taiga-NoteCommit") R, derived via the restated pasta
+     * group hash and pinned byte-for-byte against the reference's R_U/R_Z
+     * window tables (tests/test_fixed_base_tables.py). Standard limbs. */
+    static const uint64_t RX[4] = {0x8802ca95558f33acULL, 0x18eaf0144b63a217ULL,
+                                   0xae99b6368e616ee5ULL, 0x108bffda6aff53a7ULL};
+    static const uint64_t RY[4] = {0x8c0b30657333c5e1ULL, 0xb8b4f7c83326a380ULL,
+                                   0x3ebb783b59d92092ULL, 0x1319b788fe5fec16ULL};
     pt_aff a;
-    fd_limbs one;
-    fd_one_mont(one, FP);
-    fd_limbs zero = {0, 0, 0, 0};
-    fd_sub(a.x, zero, one, FP);
-    fd_add(a.y, one, one, FP);
+    fd_to_mont(a.x, RX, FP);
+    fd_to_mont(a.y, RY, FP);
     a.inf = 0;
     pt_from_aff(g, &a, FP);
 }

@@ -37,11 +37,19 @@ using PallasJac = Jac<FpCfg>;
 static const char BS_PERSONAL[17] = "Taiga_RedPallasH";
 
 inline PallasJac pallas_basepoint() {
-  // Pallas generator (-1, 2) (pasta_curves convention), Mont form
+  // RESOURCE_COMMIT_DOMAIN.R() (constant.rs:160) — the sinsemilla
+  // CommitDomain("Taiga-NoteCommit") R point, derived in-repo via the
+  // restated pasta group-hash chain and PINNED byte-for-byte against the
+  // reference's R_U/R_Z window tables (tests/test_fixed_base_tables.py).
+  // Compressed: ac338f5595ca028817a2634b14f0ea18e56e618e36b699aea753ff6adaff8b90
+  // Standard-form limbs; converted to Montgomery here.
+  static const Fp RX = {{0x8802ca95558f33acULL, 0x18eaf0144b63a217ULL,
+                         0xae99b6368e616ee5ULL, 0x108bffda6aff53a7ULL}};
+  static const Fp RY = {{0x8c0b30657333c5e1ULL, 0xb8b4f7c83326a380ULL,
+                         0x3ebb783b59d92092ULL, 0x1319b788fe5fec16ULL}};
   PallasAff g;
-  Fp one = fd_one_mont<FpCfg>();
-  g.x = fd_sub(fd_zero<FpCfg>(), one);
-  g.y = fd_add(one, one);
+  g.x = fd_to_mont(RX);
+  g.y = fd_to_mont(RY);
   return jac_from_aff(g);
 }
 
