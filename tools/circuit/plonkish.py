@@ -498,6 +498,11 @@ class Region:
         # defer to cs at close-time
         self.cs._pending_eq.append((a, b))
 
+    def constrain_constant(self, cell: Cell, cval: int):
+        """region.constrain_constant: copy-constrain a cell against a value
+        in the constants fixed column."""
+        self.cs._pending_const.append((cell, cval % F.P))
+
     def assign_advice_from_instance(self, icol, irow, col, offset) -> Cell:
         v = self.cs.instance_v[irow]
         c = Cell(col, None, v)
@@ -529,6 +534,7 @@ class ConstraintSystem:
         self.instance_v = None
         self.regions = 0
         self._pending_eq = []
+        self._pending_const = []
         self.table_rows = {}  # table col -> n assigned rows
 
     # -- configure-phase API
@@ -600,7 +606,7 @@ class ConstraintSystem:
         self.table_rows[ci] = len(values)
 
     def constrain_instance(self, cell: Cell, icol, irow):
-        self.copies.append((("instance", icol.index, irow), (cell.col, cell.row)))
+        self.copies.append(((icol, irow), (cell.col, cell.row)))
         # value check happens in mock verify
 
     def _close_region(self, r: Region):
@@ -634,7 +640,7 @@ class ConstraintSystem:
                 self.copies.append(((src.col, src.row), (col, row)))
             elif isinstance(src, tuple) and src and src[0] == "inst":
                 self.copies.append(
-                    (("instance", 0, src[1]), (col, row))
+                    ((self.instance_cols[0], src[1]), (col, row))
                 )
             if cval is not None:
                 # halo2 assigns the constant into the constants fixed column
@@ -653,6 +659,11 @@ class ConstraintSystem:
         for a, b in self._pending_eq:
             self.copies.append(((a.col, a.row), (b.col, b.row)))
         self._pending_eq = []
+        for cell, cval in self._pending_const:
+            crow = self._alloc_const_row()
+            self.fixed_vals[self.const_col.index][crow] = cval
+            self.copies.append(((self.const_col, crow), (cell.col, cell.row)))
+        self._pending_const = []
 
     def _alloc_const_row(self):
         key = self.const_col
