@@ -112,6 +112,31 @@ int tg_select_key(tg_ctx* ctx, int slot);
 int tg_create_proof(tg_ctx* ctx, const uint8_t inst_seed[32],
                     const uint8_t wit_seed[32], const uint8_t rng_seed[32],
                     uint8_t* proof_out, size_t cap, size_t* out_len);
+/* ---- drop-in circuit proving (round 2: the EXACT compliance and
+ * TrivialRL circuits; tools/circuit generates the TGD1 desc for tg_keygen
+ * and the TGW1 witness-synthesis program attached here) ----
+ * tg_witness_program_load: attach the TGW1 blob to the ACTIVE key.
+ * tg_compliance_prove: replaces ComplianceInfo::build + Proof::create
+ *   (compliance.rs:190-233, proof.rs:25-42): input = borsh ComplianceInfo
+ *   (input_resource 202B ‖ merkle path 4+33x32 ‖ anchor 32 ‖
+ *   output_resource 202B ‖ rseed 32); witness synthesized via the program;
+ *   instance_out = the 9 public-input rows (to_instance order).
+ * tg_rl_prove: TrivialResourceLogicCircuit::get_verifying_info
+ *   (resource_logic_examples.rs:116-134): witness = borsh
+ *   ResourceExistenceWitness (202 + 33x4); pad_rseed draws the 16 random
+ *   padding rows; instance_out = 22 rows.
+ * tg_witness_synthesize: advice export for parity tests (kind 0/1). */
+int tg_witness_program_load(tg_ctx* ctx, const uint8_t* tgw, size_t len);
+int tg_compliance_prove(tg_ctx* ctx, const uint8_t* info_borsh, size_t len,
+                        const uint8_t rng_seed[32], uint8_t* proof_out,
+                        size_t cap, size_t* out_len, uint8_t instance_out[288]);
+int tg_rl_prove(tg_ctx* ctx, const uint8_t* witness_borsh, size_t len,
+                const uint8_t pad_rseed[32], const uint8_t rng_seed[32],
+                uint8_t* proof_out, size_t cap, size_t* out_len,
+                uint8_t instance_out[704]);
+int tg_witness_synthesize(tg_ctx* ctx, int kind, const uint8_t* borsh, size_t len,
+                          const uint8_t pad_rseed[32], uint8_t* advice_out,
+                          uint8_t* instance_out);
 /* raw-witness variant: instance = n_instance_rows x 32B canonical reprs,
  * advice = n_advice x 2^k x 32B canonical column-major (rows beyond
  * usable are replaced by prover blinding). Same proof bytes as the seeded

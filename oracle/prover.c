@@ -198,6 +198,13 @@ static Desc* desc_parse(const uint8_t* blob, size_t len) {
     d->n = 1L << d->k;
     d->ext_n = 1L << d->ext_k;
     d->usable = d->n - (d->bf + 1);
+    /* capacity guards (must stay <= the static working arrays below;
+     * raised for the exact compliance/RL circuits: 95+ gates, 17-21
+     * fixed columns, 16 quotient pieces) */
+    if (d->n_gates > 256 || d->n_fixed > 64 || d->n_advice > 16 ||
+        d->n_perm > 32 || d->n_lookups > 4 || d->n_advice_q > 64 ||
+        d->n_fixed_q > 64 || d->n_instance_q > 8 ||
+        (d->ext_n / d->n) > 32 || d->n_instance > 1) return NULL;
     d->consts = (fd_limbs*)xmalloc(sizeof(fd_limbs) * (d->n_consts ? d->n_consts : 1));
     for (int i = 0; i < d->n_consts; i++) {
         if (fd_from_bytes(d->consts[i], rd(&p, 32), FP)) return NULL;
@@ -1329,7 +1336,7 @@ static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_
 #pragma omp parallel for schedule(static)
 #endif
         for (long i = 0; i < ext_n; i++) {
-            fd_limbs gate_vals[16];
+            fd_limbs gate_vals[256];
             for (int g = 0; g < d->n_gates; g++) expr_eval_row(gate_vals[g], &d->gates[g], &ec, i);
             fd_limbs perm_colvals[32], sigma_vals[32];
             for (int jj = 0; jj < d->n_perm; jj++) {
@@ -1408,7 +1415,7 @@ static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_
         fprintf(stderr, "dbg: h_coeff trailing zeros = %ld (need >= %d)\n", z, d->bf + 4);
     }
     int npieces = (int)(ext_n / n);
-    fd_limbs h_blind[16];
+    fd_limbs h_blind[32];
     for (int pce = 0; pce < npieces; pce++) prng_field(&rng, h_blind[pce]);
     for (int pce = 0; pce < npieces; pce++) {
         pt_aff cm;
@@ -1427,7 +1434,7 @@ static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_
     }
 
     /* 8. evals (write order defined in DESIGN.md) */
-    fd_limbs adv_eval[32], fix_eval[32], sig_eval[32], rand_eval;
+    fd_limbs adv_eval[64], fix_eval[64], sig_eval[64], rand_eval;
     fd_limbs pz_eval[8][2], pz_last_eval[8], lk_eval[4][5];
     for (int q = 0; q < d->n_advice_q; q++) {
         fd_limbs pt;
@@ -1507,9 +1514,9 @@ static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_
     }
 
     /* 9. multiopen query list (order defined in DESIGN.md) */
-    MPoly polys[128];
+    MPoly polys[192];
     int n_polys = 0;
-    MQuery queries[256];
+    MQuery queries[512];
     int n_queries = 0;
 #define ADD_POLY(COEFF, BLIND) \
     (polys[n_polys].coeff = (COEFF), fd_copy(polys[n_polys].blind, (BLIND)), n_polys++)
@@ -1527,7 +1534,7 @@ static int orc_prove_core(Pk* pk, const fd_limbs* inst_in, fd_limbs* const* adv_
         pid_lap[l] = ADD_POLY(lkAp_coeff[l], lkAp_blind[l]);
         pid_lsp[l] = ADD_POLY(lkSp_coeff[l], lkSp_blind[l]);
     }
-    int pid_fix[16];
+    int pid_fix[64];
     for (int c = 0; c < d->n_fixed; c++) pid_fix[c] = ADD_POLY(pk->fixed_coeff[c], one);
     int pid_sig[32];
     for (int j = 0; j < d->n_perm; j++) pid_sig[j] = ADD_POLY(pk->sigma_coeff[j], one);
@@ -1613,13 +1620,13 @@ int orc_multiopen_prove(const Pk* pk, tg_transcript* ts, ProofRng* rng,
     orc_ts_squeeze(ts, x2);
 
     /* group queries: per poly (in first-appearance order), its point list */
-    int order[128], n_order = 0;      /* poly ids in first appearance order */
-    int poly_set[128];                /* set index per poly */
-    fd_limbs pts[128][4];             /* per poly: its points */
-    fd_limbs evs[128][4];
-    int npts[128];
+    int order[192], n_order = 0;      /* poly ids in first appearance order */
+    int poly_set[192];                /* set index per poly */
+    fd_limbs pts[192][4];             /* per poly: its points */
+    fd_limbs evs[192][4];
+    int npts[192];
     memset(npts, 0, sizeof(npts));
-    int seen[128];
+    int seen[192];
     memset(seen, 0, sizeof(seen));
     for (int q = 0; q < n_queries; q++) {
         int pid = queries[q].poly_id;
@@ -2016,7 +2023,7 @@ static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof
     if (orc_ts_read_point(&ts, &rand_cm)) return -106;
     fd_limbs ych;
     orc_ts_squeeze(&ts, ych);
-    pt_aff h_cm[16];
+    pt_aff h_cm[32];
     for (int pce = 0; pce < npieces; pce++)
         if (orc_ts_read_point(&ts, &h_cm[pce])) return -107;
     fd_limbs x;
@@ -2029,7 +2036,7 @@ static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof
         fd_copy(xn, acc);
     }
 
-    fd_limbs adv_eval[32], fix_eval[32], sig_eval[32], rand_eval;
+    fd_limbs adv_eval[64], fix_eval[64], sig_eval[64], rand_eval;
     fd_limbs pz_eval[8][2], pz_last_eval[8], lk_eval[4][5];
     for (int q = 0; q < d->n_advice_q; q++)
         if (orc_ts_read_scalar(&ts, adv_eval[q])) return -110;
@@ -2063,7 +2070,7 @@ static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof
     {
         ScalarCtx sc = {d, (const fd_limbs*)adv_eval, (const fd_limbs*)fix_eval,
                         (const fd_limbs*)inst_eval};
-        fd_limbs gate_vals[16];
+        fd_limbs gate_vals[256];
         for (int g = 0; g < d->n_gates; g++)
             if (expr_eval_scalar(gate_vals[g], &d->gates[g], &sc)) return -120;
         fd_limbs perm_colvals[32], sigma_vals[32];
@@ -2165,7 +2172,7 @@ static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof
 
     pt_aff pcm[128];
     int n_polys = 0;
-    MQuery queries[256];
+    MQuery queries[512];
     int n_queries = 0;
 #define ADDC(CM) (pcm[n_polys] = (CM), n_polys++)
     /* order must mirror orc_prove exactly */
@@ -2183,7 +2190,7 @@ static int orc_verify_core(Pk* pk, const fd_limbs* inst_in, const uint8_t* proof
     /* fixed + sigma commitments from pk (keygen) */
     extern pt_aff* orc_pk_fixed_commits(Pk * pk);
     pt_aff* fixed_cms = orc_pk_fixed_commits(pk);
-    int pid_fix[16];
+    int pid_fix[64];
     for (int c = 0; c < d->n_fixed; c++) pid_fix[c] = ADDC(fixed_cms[c]);
     int pid_sig[32];
     for (int j = 0; j < d->n_perm; j++) pid_sig[j] = ADDC(pk->sigma_commits[j]);

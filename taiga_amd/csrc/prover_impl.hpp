@@ -70,6 +70,13 @@ inline bool pdesc_parse(PDesc& d, const uint8_t* blob, size_t len) {
   d.n = 1L << d.k;
   d.ext_n = 1L << d.ext_k;
   d.usable = d.n - (d.bf + 1);
+  // capacity guards, matched to the static device-side arrays
+  // (HFoldArgs in prover_gpu.inc) and host stacks; sized for the exact
+  // compliance/RL circuits (95+ gates, 17-21 fixed columns, ext 2^19)
+  if (d.n_gates > 256 || d.n_fixed > 64 || d.n_advice > 16 || d.n_perm > 32 ||
+      d.n_lookups > 4 || d.n_advice_q > 64 || d.n_fixed_q > 64 ||
+      d.n_instance_q > 2 || d.n_instance > 1 || (d.ext_n / d.n) > 32)
+    return false;
   auto rd_fp = [&]() {
     Fp v;
     memcpy(v.l, p, 32);

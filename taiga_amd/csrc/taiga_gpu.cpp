@@ -20,6 +20,7 @@
 #include "binding_sig.hpp"
 #include "tx_wire.hpp"
 #include "fd28.hpp"
+#include "witness.hpp"
 #include "prover_impl.hpp"
 
 namespace taiga {
@@ -979,6 +980,112 @@ extern "C" {
  * n_instance_rows x 32B canonical reprs; advice = n_advice x 2^k x 32B
  * canonical, column-major (rows beyond usable = n-(bf+1) are replaced by
  * blinding). */
+/* attach a TGW1 witness-synthesis program (tools/circuit/emit.py) to the
+ * ACTIVE key slot; required by tg_compliance_prove / tg_rl_prove */
+int tg_witness_program_load(tg_ctx* ctx, const uint8_t* tgw, size_t len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  if (!c->ppk || !c->ppk->ready) return TG_ERR_STATE;
+  if (!c->ppk->tgw.parse(tgw, len)) return TG_ERR_BADARG;
+  if ((int)c->ppk->tgw.k != c->ppk->d.k) return TG_ERR_BADARG;
+  return TG_OK;
+}
+
+/* drop-in ComplianceInfo prove (replaces ComplianceInfo::build +
+ * Proof::create — compliance.rs:190-233 + proof.rs:25-42): parses the
+ * borsh ComplianceInfo, synthesizes the exact compliance circuit witness
+ * via the attached program, reads the circuit-computed public inputs back
+ * (nf, cm, delta, RL commitments; anchor from the blob), proves on the
+ * GPU. instance_out = 9 x 32B rows (CompliancePublicInputs::to_instance
+ * order). */
+int tg_compliance_prove(tg_ctx* ctx, const uint8_t* info_borsh, size_t len,
+                        const uint8_t rng_seed[32], uint8_t* proof_out,
+                        size_t cap, size_t* out_len, uint8_t instance_out[288]) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  if (!c->ppk || !c->ppk->ready || !c->ppk->tgw.ready) return TG_ERR_STATE;
+  PDesc& d = c->ppk->d;
+  if (d.n_instance_rows != 9 || c->ppk->tgw.n_inputs != 124) return TG_ERR_STATE;
+  std::vector<Fp> inputs;
+  if (!compliance_inputs(info_borsh, len, inputs)) return TG_ERR_ENCODING;
+  std::vector<uint8_t> advice(32ul * d.n_advice * d.n, 0);
+  if (!c->ppk->tgw.run(inputs, d.n_advice, advice.data())) return TG_ERR_BADARG;
+  uint8_t inst[288];
+  memset(inst, 0, sizeof(inst));
+  {  // anchor row from the blob (instance row 1 = input slot 1)
+    Fp a = fd_from_mont(inputs[1]);
+    memcpy(inst + 32, a.l, 32);
+  }
+  c->ppk->tgw.read_instance(d.n_advice, advice.data(), inst);
+  std::vector<uint8_t> proof;
+  int rc = pprove_raw(c, *c->ppk, inst, advice.data(), rng_seed, proof);
+  if (rc != 0) return rc;
+  if (proof.size() > cap) return TG_ERR_BADARG;
+  memcpy(proof_out, proof.data(), proof.size());
+  *out_len = proof.size();
+  if (instance_out) memcpy(instance_out, inst, 288);
+  return TG_OK;
+}
+
+/* drop-in TrivialRL prove (resource_logic_examples.rs get_verifying_info):
+ * witness = borsh ResourceExistenceWitness; pad_rseed drives the random
+ * instance padding (rows 6..21). instance_out = 22 x 32B. */
+int tg_rl_prove(tg_ctx* ctx, const uint8_t* witness_borsh, size_t len,
+                const uint8_t pad_rseed[32], const uint8_t rng_seed[32],
+                uint8_t* proof_out, size_t cap, size_t* out_len,
+                uint8_t instance_out[704]) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  if (!c->ppk || !c->ppk->ready || !c->ppk->tgw.ready) return TG_ERR_STATE;
+  PDesc& d = c->ppk->d;
+  if (d.n_instance_rows != 22 || c->ppk->tgw.n_inputs != 41) return TG_ERR_STATE;
+  std::vector<Fp> inputs;
+  uint8_t inst[704];
+  memset(inst, 0, sizeof(inst));
+  if (!rl_inputs(witness_borsh, len, pad_rseed, inputs, inst + 6 * 32))
+    return TG_ERR_ENCODING;
+  std::vector<uint8_t> advice(32ul * d.n_advice * d.n, 0);
+  if (!c->ppk->tgw.run(inputs, d.n_advice, advice.data())) return TG_ERR_BADARG;
+  c->ppk->tgw.read_instance(d.n_advice, advice.data(), inst);
+  std::vector<uint8_t> proof;
+  int rc = pprove_raw(c, *c->ppk, inst, advice.data(), rng_seed, proof);
+  if (rc != 0) return rc;
+  if (proof.size() > cap) return TG_ERR_BADARG;
+  memcpy(proof_out, proof.data(), proof.size());
+  *out_len = proof.size();
+  if (instance_out) memcpy(instance_out, inst, 704);
+  return TG_OK;
+}
+
+/* synthesized-advice export for parity tests: runs the attached program on
+ * a borsh witness (kind 0 = compliance, 1 = RL) and writes the advice
+ * column bytes (n_advice x 2^k x 32). */
+int tg_witness_synthesize(tg_ctx* ctx, int kind, const uint8_t* borsh, size_t len,
+                          const uint8_t pad_rseed[32], uint8_t* advice_out,
+                          uint8_t* instance_out) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  if (!c->ppk || !c->ppk->ready || !c->ppk->tgw.ready) return TG_ERR_STATE;
+  PDesc& d = c->ppk->d;
+  std::vector<Fp> inputs;
+  size_t ninst = (size_t)d.n_instance_rows * 32;
+  std::vector<uint8_t> inst(ninst, 0);
+  if (kind == 0) {
+    if (!compliance_inputs(borsh, len, inputs)) return TG_ERR_ENCODING;
+    Fp a = fd_from_mont(inputs[1]);
+    memcpy(inst.data() + 32, a.l, 32);
+  } else {
+    if (ninst < 22 * 32) return TG_ERR_STATE;
+    if (!rl_inputs(borsh, len, pad_rseed, inputs, inst.data() + 6 * 32))
+      return TG_ERR_ENCODING;
+  }
+  memset(advice_out, 0, 32ul * d.n_advice * d.n);
+  if (!c->ppk->tgw.run(inputs, d.n_advice, advice_out)) return TG_ERR_BADARG;
+  c->ppk->tgw.read_instance(d.n_advice, advice_out, inst.data());
+  if (instance_out) memcpy(instance_out, inst.data(), ninst);
+  return TG_OK;
+}
+
 int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* advice,
                         const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
                         size_t* out_len) {

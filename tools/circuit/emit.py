@@ -1,23 +1,27 @@
 """Finalize a synthesized circuit: selector compression, mock verification,
 TGD2 circuit-description emission and TGW1 witness-program emission.
 
-TGD2 blob = TGD1 (tools/gen_cs1.py format) with:
-  - magic "TGD2"
-  - an extra per-gate gating hint after the gates section:
-    n_gates x (u32 gate_col /* fixed col idx */, i32 rot) — the h-fold
-    kernels read the gating fixed value first and skip the expression when
-    it is zero (the gate poly is sel*content, so the result is identical).
-  - capacities beyond CS1's (checked by the parsers).
+The desc blob uses the TGD1 format (tools/gen_cs1.py) unchanged — the
+real circuits just use larger counts (95 constraints, 17+ fixed columns,
+postfix stack depth up to 8), which the provers' capacities cover.
+(Selector-gated skipping is NOT possible in the quotient evaluation: on
+the extended coset the selector polynomials are dense.)
 
 TGW1 blob (witness-synthesis program, interpreted by oracle/witness.c and
 taiga_amd/csrc/witness.hpp at prove time):
-  magic "TGW1", u32: n_inputs, n_consts, n_ops, n_stores, n_advice, k
+  magic "TGW1", u32: n_inputs, n_consts, n_ops, n_stores, n_expose, k
   consts:  n_consts x 32B canonical Fp
-  ops:     n_ops x { u8 opcode, u8 pad, u16 b, u32 a }  (8 B)
-           opcodes: 0 LOADI(a=input) 1 CONST(a=idx) 2 ADD(a,b=regs...)
-           — see plonkish.py W_* enums; a/b are reg indices except
+  ops:     n_ops x { u8 opcode, u8 pad, u16 pad2, u32 a, u32 b }  (12 B)
+           opcodes 0..9 = LOADI CONST ADD SUB MUL INV0 NEG SQRT0 BIT BYTE
+           (plonkish.py W_* enums); a/b are reg indices except
            LOADI/CONST (a = input/const index) and BIT/BYTE (b = position)
-  stores:  n_stores x { u32 col, u32 row, u32 reg }
+  stores:  n_stores x { u32 col, u32 row, u32 reg }   (advice cells)
+  expose:  n_expose x { u32 instance_row, u32 col, u32 row } — instance
+           rows whose value the circuit computes (a copy constraint ties
+           the instance row to advice (col,row)); the prove-time builder
+           reads them from the synthesized advice instead of re-deriving
+           the public inputs host-side. Rows not listed (the compliance
+           anchor, the RL random padding) come from the witness blob.
 Register i is defined by op i (SSA; single pass).
 """
 from __future__ import annotations
@@ -275,7 +279,7 @@ def emit_desc(fin: Finalized):
             arr[row * 32:(row + 1) * 32] = (v % F.P).to_bytes(32, "little")
         fixed_dense.append(bytes(arr))
 
-    out = b"TGD2"
+    out = b"TGD1"
     out += struct.pack(
         "<15I", cs.k, fin.ext_k, len(cs.fixed_cols), len(cs.advice_cols),
         len(cs.instance_cols), fin.bf, len(gate_ops), len(perm_cols),
@@ -297,9 +301,6 @@ def emit_desc(fin: Finalized):
         out += struct.pack("<I", len(ops))
         for tag, a, b in ops:
             out += struct.pack("<IIi", tag, a, b)
-    # TGD2 extension: gating hints
-    for gcol, grot in gate_gating:
-        out += struct.pack("<Ii", gcol, grot)
     for ins, tabs in lk_enc:
         out += struct.pack("<II", len(ins), len(tabs))
         for ops in ins + tabs:
@@ -326,13 +327,22 @@ def emit_witness_program(cs: ConstraintSystem):
     for ci, colvals in enumerate(cs.advice_vals):
         for row, reg in sorted(colvals.items()):
             stores.append((ci, row, reg.r))
+    expose = {}
+    for (c1, r1), (c2, r2) in cs.copies:
+        if c1.kind == "instance" and c2.kind == "advice" and r1 not in expose:
+            expose[r1] = (c2.index, r2)
+        elif c2.kind == "instance" and c1.kind == "advice" and r2 not in expose:
+            expose[r2] = (c1.index, r1)
     out = b"TGW1"
     out += struct.pack("<6I", prog.n_inputs, len(prog.consts), len(prog.ops),
-                       len(stores), len(cs.advice_cols), cs.k)
+                       len(stores), len(expose), cs.k)
     for v in prog.consts:
         out += v.to_bytes(32, "little")
     for (op, a, b) in prog.ops:
         out += struct.pack("<BBHII", op, 0, 0, a & 0xFFFFFFFF, b & 0xFFFFFFFF)
     for (col, row, reg) in stores:
         out += struct.pack("<III", col, row, reg)
+    for irow in sorted(expose):
+        col, row = expose[irow]
+        out += struct.pack("<III", irow, col, row)
     return out

@@ -48,11 +48,24 @@ def sample_compliance_inputs(tag=b"sample0"):
     path = [(fp(b"node%d" % i), bool(det(tag + b"lr%d" % i, 2)))
             for i in range(32)]
     anchor = hc.merkle_root(input_res.commitment(), path)
-    rcv = det(tag + b"rcv", F.Q).to_bytes(32, "little")
-    r_in = fp(b"rlcmr_in")
-    r_out = fp(b"rlcmr_out")
-    return build_compliance_inputs(input_res, path, anchor, output_res, rcv,
-                                   r_in, r_out)
+    # rseed-driven randomness exactly like ComplianceInfo (compliance.rs):
+    rseed = hashlib.blake2b(tag + b"rseed", digest_size=32).digest()
+    rcv_int = hc.prf_expand_field(rseed, hc.PRF_EXPAND_VCM_R, F.Q)
+    rcv = rcv_int.to_bytes(32, "little")
+    r_in = hc.prf_expand_field(rseed, hc.PRF_EXPAND_INPUT_RESOURCE_LOGIC_CM_R)
+    r_out = hc.prf_expand_field(rseed, hc.PRF_EXPAND_OUTPUT_RESOURCE_LOGIC_CM_R)
+    instance, inputs = build_compliance_inputs(
+        input_res, path, anchor, output_res, rcv, r_in, r_out)
+    # borsh ComplianceInfo (compliance.rs:51-59)
+    import struct as _s
+    borsh = input_res.borsh()
+    borsh += _s.pack("<I", 32)
+    for node, is_left in path:
+        borsh += F.to_repr(node) + bytes([1 if is_left else 0])
+    borsh += F.to_repr(anchor)
+    borsh += output_res.borsh()
+    borsh += rseed
+    return instance, inputs, borsh
 
 
 def sample_rl_inputs(tag=b"rlsample0"):
@@ -65,14 +78,22 @@ def sample_rl_inputs(tag=b"rlsample0"):
             for i in range(4)]
     # is_input must match !path[0].is_left (resource_tree.rs is_input())
     path[0] = (path[0][0], False)
-    padding = [fp(b"pad%d" % i) for i in range(16)]
-    return build_rl_inputs(res, path, True, padding)
+    pad_rseed = hashlib.blake2b(tag + b"padseed", digest_size=32).digest()
+    padding = hc.random_seed_padding(pad_rseed, 16)
+    instance, inputs = build_rl_inputs(res, path, True, padding)
+    # borsh ResourceExistenceWitness (resource_tree.rs:70-81)
+    borsh = res.borsh()
+    for node, is_left in path:
+        borsh += F.to_repr(node) + bytes([1 if is_left else 0])
+    return instance, inputs, borsh, pad_rseed
 
 
 def run(model_cls, builder, name, check):
     t0 = time.time()
     model = model_cls()
-    instance, inputs = builder()
+    built = builder()
+    instance, inputs = built[0], built[1]
+    extra = built[2:]
     cs = model.synthesize(inputs)
     t1 = time.time()
     fin = emit.finalize(cs, instance, name)
@@ -112,6 +133,10 @@ def run(model_cls, builder, name, check):
         "instance": [hex(v) for v in instance],
         "advice_col_blake2b": colhash,
     }
+    if extra:
+        fix["witness_borsh"] = extra[0].hex()
+    if len(extra) > 1:
+        fix["pad_rseed"] = extra[1].hex()
     open(os.path.join(GOLDEN, f"{name}_sample.json"), "w").write(
         json.dumps(fix, indent=1))
     return cs, fin, instance
