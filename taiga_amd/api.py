@@ -283,6 +283,40 @@ class TaigaGpu:
             self._h, instance, advice, rng_seed, out, len(out), ctypes.byref(out_len)))
         return out.raw[: out_len.value]
 
+    def witness_program_load(self, tgw: bytes):
+        """Attach a TGW1 witness-synthesis program to the active key."""
+        self._ck(self._lib.tg_witness_program_load(self._h, tgw, len(tgw)))
+
+    def compliance_prove(self, info_borsh: bytes, rng_seed: bytes):
+        """Drop-in ComplianceInfo::build + Proof::create; returns
+        (proof_bytes, instance_bytes 9x32)."""
+        out = ctypes.create_string_buffer(1 << 16)
+        out_len = ctypes.c_size_t()
+        inst = ctypes.create_string_buffer(288)
+        self._ck(self._lib.tg_compliance_prove(
+            self._h, info_borsh, len(info_borsh), rng_seed, out, len(out),
+            ctypes.byref(out_len), inst))
+        return out.raw[: out_len.value], inst.raw
+
+    def rl_prove(self, witness_borsh: bytes, pad_rseed: bytes, rng_seed: bytes):
+        """Drop-in TrivialRL get_verifying_info; returns
+        (proof_bytes, instance_bytes 22x32)."""
+        out = ctypes.create_string_buffer(1 << 16)
+        out_len = ctypes.c_size_t()
+        inst = ctypes.create_string_buffer(704)
+        self._ck(self._lib.tg_rl_prove(
+            self._h, witness_borsh, len(witness_borsh), pad_rseed, rng_seed,
+            out, len(out), ctypes.byref(out_len), inst))
+        return out.raw[: out_len.value], inst.raw
+
+    def witness_synthesize(self, kind: int, borsh: bytes, pad_rseed: bytes,
+                           n_advice: int, n: int, n_inst_rows: int):
+        adv = ctypes.create_string_buffer(32 * n_advice * n)
+        inst = ctypes.create_string_buffer(32 * n_inst_rows)
+        self._ck(self._lib.tg_witness_synthesize(
+            self._h, kind, borsh, len(borsh), pad_rseed, adv, inst))
+        return adv.raw, inst.raw
+
     def verify_proof(self, inst_seed: bytes, proof: bytes) -> bool:
         rc = self._lib.tg_verify_proof(self._h, inst_seed, proof, len(proof))
         if rc == 0:

@@ -16,12 +16,15 @@ def lib():
     global _lib
     if _lib is None:
         _lib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
-        desc = open(os.path.join(GOLDEN, "cs1.desc"), "rb").read()
-        srs = open(os.path.join(GOLDEN, "params_15"), "rb").read()
-        rc = _lib.orc_prover_init(desc, len(desc), srs, len(srs))
-        assert rc in (0, 1)
         _lib.orc_prove_cs1.restype = ctypes.c_long
         _lib.orc_dbg_perm_base.restype = ctypes.c_long
+    # always (re)initialize with the CS1 desc: liboracle holds ONE global
+    # PK and other test modules init it with the compliance/RL descs
+    _lib.orc_prover_reset()
+    desc = open(os.path.join(GOLDEN, "cs1.desc"), "rb").read()
+    srs = open(os.path.join(GOLDEN, "params_15"), "rb").read()
+    rc = _lib.orc_prover_init(desc, len(desc), srs, len(srs))
+    assert rc in (0, 1)
     return _lib
 
 

@@ -20,6 +20,7 @@ def oracle_pk():
     lib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
     desc = open(os.path.join(GOLDEN, "cs1.desc"), "rb").read()
     srs = open(os.path.join(GOLDEN, "params_15"), "rb").read()
+    lib.orc_prover_reset()  # other modules init the global PK with other descs
     rc = lib.orc_prover_init(desc, len(desc), srs, len(srs))
     assert rc in (0, 1)
     lib.orc_prove_cs1.restype = ctypes.c_long
