@@ -1086,6 +1086,273 @@ int tg_witness_synthesize(tg_ctx* ctx, int kind, const uint8_t* borsh, size_t le
   return TG_OK;
 }
 
+/* ---- ShieldedPartialTransaction::build / verify (shielded_ptx.rs:98-137,
+ * BASELINE configs[3]) ----
+ * Builds one borsh ShieldedPartialTransaction: n_compliance compliance
+ * proofs (borsh ComplianceInfo units, 1528 B each; key slot
+ * slot_compliance) + per input/output resource one TrivialRL proof (borsh
+ * ResourceExistenceWitness units, 334 B each; slot_rl), binding_sig_r =
+ * sum of the units' rcv (Some), empty hints. Per-proof randomness is
+ * drawn from ONE ChaCha20 stream over rng_seed in build order:
+ * compliance proof seeds first, then per RL (pad_rseed, proof seed) —
+ * the determinized stand-in for the reference's single &mut rng
+ * (shielded_ptx.rs:105-125; DESIGN.md randomness ledger). */
+int tg_ptx_build(tg_ctx* ctx, int slot_compliance, int slot_rl,
+                 uint32_t n_compliance, const uint8_t* compliance_units,
+                 uint32_t n_in, uint32_t n_out, const uint8_t* rl_units,
+                 const uint8_t rng_seed[32], uint8_t* ptx_out, size_t cap,
+                 size_t* out_len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  if (n_compliance > 64 || n_in + n_out > 128) return TG_ERR_BADARG;
+  Drbg stream(rng_seed);
+  std::vector<uint8_t> out;
+  auto put_u32 = [&](uint32_t v) {
+    out.push_back((uint8_t)v);
+    out.push_back((uint8_t)(v >> 8));
+    out.push_back((uint8_t)(v >> 16));
+    out.push_back((uint8_t)(v >> 24));
+  };
+  // compliances
+  put_u32(n_compliance);
+  Fq rcv_sum = fd_zero<FqCfg>();
+  int rc = tg_select_key(ctx, slot_compliance);
+  if (rc) return rc;
+  for (uint32_t i = 0; i < n_compliance; i++) {
+    const uint8_t* unit = compliance_units + 1528ul * i;
+    uint8_t seed[32];
+    stream.bytes(seed, 32);
+    uint8_t proof[1 << 14];
+    size_t plen = 0;
+    uint8_t inst[288];
+    rc = tg_compliance_prove(ctx, unit, 1528, seed, proof, sizeof(proof), &plen,
+                             inst);
+    if (rc) return rc;
+    // rcv from the unit's rseed (last 32 bytes)
+    rcv_sum = fd_add(rcv_sum, prf_expand<FqCfg>(unit + 1528 - 32, 3));
+    put_u32((uint32_t)plen);
+    out.insert(out.end(), proof, proof + plen);
+    // CompliancePublicInputs borsh (compliance.rs:82-93):
+    // anchor ‖ nf ‖ cm ‖ delta(compressed) ‖ rlcm_in ‖ rlcm_out
+    out.insert(out.end(), inst + 32, inst + 64);   // anchor (row 1)
+    out.insert(out.end(), inst, inst + 32);        // nf (row 0)
+    out.insert(out.end(), inst + 64, inst + 96);   // cm (row 2)
+    {  // delta rows 3,4 -> compressed pallas point
+      Fp x, y;
+      memcpy(x.l, inst + 96, 32);
+      memcpy(y.l, inst + 128, 32);
+      uint8_t comp[32];
+      if (fd_is_zero(x) && fd_is_zero(y)) {
+        memset(comp, 0, 32);
+      } else {
+        memcpy(comp, x.l, 32);
+        comp[31] |= (uint8_t)((y.l[0] & 1) << 7);
+      }
+      out.insert(out.end(), comp, comp + 32);
+    }
+    for (int half = 0; half < 2; half++) {  // rlcm halves -> 32B commitment
+      uint8_t cmb[32];
+      memcpy(cmb, inst + (160 + 64 * half), 16);
+      memcpy(cmb + 16, inst + (192 + 64 * half), 16);
+      out.insert(out.end(), cmb, cmb + 32);
+    }
+  }
+  // RL sets (inputs then outputs)
+  rc = tg_select_key(ctx, slot_rl);
+  if (rc) return rc;
+  PPk* rlk = c->ppk;
+  std::vector<uint8_t> vk_bytes;
+  for (const VestaAff& p : rlk->fixed_commits) {
+    uint8_t b[32];
+    Transcript::compress(b, p);
+    vk_bytes.insert(vk_bytes.end(), b, b + 32);
+  }
+  for (const VestaAff& p : rlk->sigma_commits) {
+    uint8_t b[32];
+    Transcript::compress(b, p);
+    vk_bytes.insert(vk_bytes.end(), b, b + 32);
+  }
+  for (int grp = 0; grp < 2; grp++) {
+    uint32_t cnt = grp == 0 ? n_in : n_out;
+    uint32_t base = grp == 0 ? 0 : n_in;
+    put_u32(cnt);
+    for (uint32_t i = 0; i < cnt; i++) {
+      const uint8_t* unit = rl_units + 334ul * (base + i);
+      uint8_t pad[32], seed[32];
+      stream.bytes(pad, 32);
+      stream.bytes(seed, 32);
+      uint8_t proof[1 << 14];
+      size_t plen = 0;
+      uint8_t inst[704];
+      rc = tg_rl_prove(ctx, unit, 334, pad, seed, proof, sizeof(proof), &plen,
+                       inst);
+      if (rc) return rc;
+      out.insert(out.end(), vk_bytes.begin(), vk_bytes.end());
+      put_u32((uint32_t)plen);
+      out.insert(out.end(), proof, proof + plen);
+      out.insert(out.end(), inst, inst + 704);
+      put_u32(0);  // app_dynamic_resource_logic_verifying_info: empty vec
+    }
+  }
+  out.push_back(1);  // Some(binding_sig_r)
+  {
+    Fq s = fd_from_mont(rcv_sum);
+    uint8_t b[32];
+    memcpy(b, s.l, 32);
+    out.insert(out.end(), b, b + 32);
+  }
+  put_u32(0);  // hints: empty
+  if (out.size() > cap) return TG_ERR_BADARG;
+  memcpy(ptx_out, out.data(), out.size());
+  *out_len = out.size();
+  return TG_OK;
+}
+
+/* verify one borsh ShieldedPartialTransaction end-to-end
+ * (ShieldedPartialTransaction::execute, shielded_ptx.rs:232-240):
+ * batch-verify ALL compliance proofs (slot_compliance) and ALL RL proofs
+ * (slot_rl, vk bytes must match the slot's key), then the consistency
+ * checks: every RL root equals the others (resource merkle root), input
+ * RL self-ids == compliance nfs, output RL self-ids == compliance cms.
+ * Returns TG_OK, -1 proof failure, -3xx consistency, -2xx structure. */
+int tg_ptx_verify(tg_ctx* ctx, int slot_compliance, int slot_rl,
+                  const uint8_t* ptx, size_t len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  int rc = tg_select_key(ctx, slot_rl);
+  if (rc) return rc;
+  PPk* rlk = c->ppk;
+  uint32_t vk_len = 32u * (uint32_t)(rlk->fixed_commits.size() +
+                                     rlk->sigma_commits.size());
+  TxCursor cur{ptx, len};
+  uint32_t n_cvi;
+  if (!cur.u32(n_cvi) || n_cvi > 64) return -201;
+  std::vector<const uint8_t*> cproof;
+  std::vector<size_t> cplen;
+  std::vector<uint8_t> cinst;  // 9 rows x 32 per proof
+  std::vector<uint8_t> nfs, cms;
+  for (uint32_t i = 0; i < n_cvi; i++) {
+    uint32_t plen;
+    const uint8_t* p;
+    if (!cur.u32(plen) || plen > (1u << 20) || !cur.take(p, plen)) return -202;
+    cproof.push_back(p);
+    cplen.push_back(plen);
+    const uint8_t* inst;
+    if (!cur.take(inst, 192)) return -203;
+    // expand the 192B borsh block into the 9 instance rows
+    // [nf, anchor, cm, dx, dy, in1, in2, out1, out2]
+    uint8_t rows[288];
+    memcpy(rows, inst + 32, 32);       // nf
+    memcpy(rows + 32, inst, 32);       // anchor
+    memcpy(rows + 64, inst + 64, 32);  // cm
+    {
+      PallasJac d;
+      bool all0 = true;
+      for (int b = 0; b < 32; b++)
+        if (inst[96 + b]) all0 = false;
+      if (all0) {
+        memset(rows + 96, 0, 64);
+      } else {
+        if (!pallas_decompress(d, inst + 96)) return -204;
+        PallasAff a = jac_to_aff(d);
+        Fp x = fd_from_mont(a.x), y = fd_from_mont(a.y);
+        memcpy(rows + 96, x.l, 32);
+        memcpy(rows + 128, y.l, 32);
+      }
+    }
+    for (int half = 0; half < 2; half++) {
+      memset(rows + 160 + 64 * half, 0, 64);
+      memcpy(rows + 160 + 64 * half, inst + 128 + 32 * half, 16);
+      memcpy(rows + 192 + 64 * half, inst + 128 + 32 * half + 16, 16);
+    }
+    cinst.insert(cinst.end(), rows, rows + 288);
+    nfs.insert(nfs.end(), rows, rows + 32);
+    cms.insert(cms.end(), rows + 64, rows + 96);
+  }
+  uint32_t n_in, n_out;
+  std::vector<const uint8_t*> rproof;
+  std::vector<size_t> rplen;
+  std::vector<const uint8_t*> rinst;
+  std::vector<uint8_t> self_ids, roots;
+  for (int grp = 0; grp < 2; grp++) {
+    uint32_t cnt;
+    if (!cur.u32(cnt) || cnt > 128) return -205;
+    (grp == 0 ? n_in : n_out) = cnt;
+    for (uint32_t i = 0; i < cnt; i++) {
+      const uint8_t* vkb;
+      if (!cur.take(vkb, vk_len)) return -206;
+      if (memcmp(vkb, [&] {
+            static thread_local std::vector<uint8_t> vk_ref;
+            vk_ref.clear();
+            for (const VestaAff& p : rlk->fixed_commits) {
+              uint8_t b[32];
+              Transcript::compress(b, p);
+              vk_ref.insert(vk_ref.end(), b, b + 32);
+            }
+            for (const VestaAff& p : rlk->sigma_commits) {
+              uint8_t b[32];
+              Transcript::compress(b, p);
+              vk_ref.insert(vk_ref.end(), b, b + 32);
+            }
+            return vk_ref.data();
+          }(), vk_len) != 0)
+        return -207;  // unknown RL vk (only the TrivialRL key is loaded)
+      uint32_t plen;
+      const uint8_t* p;
+      if (!cur.u32(plen) || plen > (1u << 20) || !cur.take(p, plen)) return -208;
+      const uint8_t* inst;
+      if (!cur.take(inst, 22 * 32)) return -209;
+      rproof.push_back(p);
+      rplen.push_back(plen);
+      rinst.push_back(inst);
+      roots.insert(roots.end(), inst, inst + 32);
+      self_ids.insert(self_ids.end(), inst + 32, inst + 64);
+      uint32_t ndyn;
+      if (!cur.u32(ndyn) || ndyn != 0) return -210;  // dynamic RLs: not built
+    }
+  }
+  const uint8_t* b;
+  if (!cur.take(b, 1)) return -211;
+  if (*b == 1 && !cur.take(b, 32)) return -212;
+  uint32_t n_hints;
+  if (!cur.u32(n_hints) || !cur.take(b, n_hints)) return -213;
+  if (cur.left != 0) return -214;
+  // consistency (shielded_ptx.rs:156-230): roots all equal; input self-ids
+  // match nfs in order; output self-ids match cms
+  for (uint32_t i = 1; i < n_in + n_out; i++)
+    if (memcmp(roots.data(), roots.data() + 32ul * i, 32) != 0) return -301;
+  if (n_in != n_cvi || n_out != n_cvi) return -302;
+  for (uint32_t i = 0; i < n_in; i++)
+    if (memcmp(self_ids.data() + 32ul * i, nfs.data() + 32ul * i, 32) != 0)
+      return -303;
+  for (uint32_t i = 0; i < n_out; i++)
+    if (memcmp(self_ids.data() + 32ul * (n_in + i), cms.data() + 32ul * i, 32) != 0)
+      return -304;
+  // batch-verify RL proofs on the RL key
+  {
+    std::vector<uint8_t> insts;
+    for (const uint8_t* ip : rinst) insts.insert(insts.end(), ip, ip + 22 * 32);
+    std::vector<uint8_t> proofs;
+    for (size_t i = 0; i < rproof.size(); i++)
+      proofs.insert(proofs.end(), rproof[i], rproof[i] + rplen[i]);
+    rc = tg_verify_batch_raw(ctx, rproof.size(), insts.data(), proofs.data(),
+                             rplen.data());
+    if (rc) return rc == TG_ERR_BADARG ? rc : -1;
+  }
+  // batch-verify compliance proofs on the compliance key
+  rc = tg_select_key(ctx, slot_compliance);
+  if (rc) return rc;
+  {
+    std::vector<uint8_t> proofs;
+    for (size_t i = 0; i < cproof.size(); i++)
+      proofs.insert(proofs.end(), cproof[i], cproof[i] + cplen[i]);
+    rc = tg_verify_batch_raw(ctx, cproof.size(), cinst.data(), proofs.data(),
+                             cplen.data());
+    if (rc) return rc == TG_ERR_BADARG ? rc : -1;
+  }
+  return TG_OK;
+}
+
 int tg_create_proof_raw(tg_ctx* ctx, const uint8_t* instance, const uint8_t* advice,
                         const uint8_t rng_seed[32], uint8_t* proof_out, size_t cap,
                         size_t* out_len) {
