@@ -2946,3 +2946,18 @@ int orc_cs1_export_witness(const uint8_t inst_seed[32], const uint8_t wit_seed[3
     free(inst);
     return 0;
 }
+
+/* vk bytes export: the restated halo2-0.3 VerifyingKey::write layout —
+ * fixed commitments then permutation commitments, 32-B compressed each
+ * (resource_logic_circuit.rs:175-188 embeds these in the wire format). */
+long orc_vk_bytes(uint8_t* out, long cap) {
+    if (!g_pk) return -1;
+    Desc* d = g_pk->d;
+    long need = 32L * (d->n_fixed + d->n_perm);
+    if (need > cap) return -2;
+    pt_aff* fixed_cms = orc_pk_fixed_commits(g_pk);
+    for (int c = 0; c < d->n_fixed; c++) pt_compress(out + 32L * c, &fixed_cms[c], &FD_Q);
+    for (int j = 0; j < d->n_perm; j++)
+        pt_compress(out + 32L * (d->n_fixed + j), &g_pk->sigma_commits[j], &FD_Q);
+    return need;
+}
