@@ -35,13 +35,10 @@ def pytest_configure(config):
 def pytest_collection_modifyitems(config, items):
     if config.getoption("-m"):
         return
-    # default runs (no -m filter) skip gpu tests when no device is present
-    try:
-        import torch
-
-        has_gpu = torch.cuda.is_available()
-    except Exception:
-        has_gpu = False
+    # default runs (no -m filter) skip gpu tests when no device is present.
+    # Probe /dev/kfd instead of importing torch: pulling in torch's OpenMP
+    # runtime next to the oracle's libgomp slows oracle proves ~4-10x.
+    has_gpu = os.path.exists("/dev/kfd")
     if has_gpu:
         return
     skip = pytest.mark.skip(reason="no GPU in this container")
