@@ -65,12 +65,80 @@ def gen_scalars(n, seed):
     return arr.tobytes()
 
 
+def _make_ptx_units(rank):
+    """Synthetic 2-in/2-out ptx witness units (seeded per rank):
+    2 borsh ComplianceInfo (1528 B) + 4 borsh ResourceExistenceWitness
+    (334 B) over the ptx resource tree [nf_1, cm_1, nf_2, cm_2]."""
+    import hashlib
+    import struct
+
+    sys.path.insert(0, os.path.join(REPO, "tools"))
+    from circuit import fields as F
+    from circuit import hostcrypto as hc
+
+    def det(seed, n):
+        return int.from_bytes(hashlib.blake2b(seed, digest_size=64).digest(),
+                              "little") % n
+
+    def mkres(tag, nk_is_key=True, nonce=None):
+        fp = lambda sfx: det(tag + sfx, F.P)
+        return hc.Resource(
+            logic=fp(b"logic"), label=fp(b"label"), value=fp(b"value"),
+            quantity=det(tag + b"q", 1 << 64), nk=fp(b"nk"),
+            nk_is_key=nk_is_key,
+            nonce=nonce if nonce is not None else fp(b"nonce"),
+            is_ephemeral=False, rseed=fp(b"rseed"))
+
+    comp_units = b""
+    leaves = []
+    resources = []
+    for i in range(2):
+        tag = b"bench-r%d-%d" % (rank, i)
+        rin = mkres(tag + b"in")
+        nf = rin.get_nf()
+        rout = mkres(tag + b"out", nonce=nf)
+        path = [(det(tag + b"n%d" % j, F.P), bool(det(tag + b"l%d" % j, 2)))
+                for j in range(32)]
+        anchor = hc.merkle_root(rin.commitment(), path)
+        rseed = hashlib.blake2b(tag + b"rs", digest_size=32).digest()
+        b = rin.borsh() + struct.pack("<I", 32)
+        for node, is_left in path:
+            b += F.to_repr(node) + bytes([1 if is_left else 0])
+        b += F.to_repr(anchor) + rout.borsh() + rseed
+        comp_units += b
+        leaves += [nf, rout.commitment()]
+        resources.append((rin, rout))
+    layer = leaves + [0] * (16 - len(leaves))
+    layers = [layer]
+    while len(layer) > 1:
+        layer = [hc.poseidon_hash(layer[i], layer[i + 1])
+                 for i in range(0, len(layer), 2)]
+        layers.append(layer)
+    rl_in = rl_out = b""
+    for i, (rin, rout) in enumerate(resources):
+        for j, res in ((0, rin), (1, rout)):
+            p = 2 * i + j
+            wb = res.borsh()
+            pp = p
+            for lvl in range(4):
+                sib = pp ^ 1
+                wb += F.to_repr(layers[lvl][sib]) + bytes([1 if sib < pp else 0])
+                pp >>= 1
+            if j == 0:
+                rl_in += wb
+            else:
+                rl_out += wb
+    return comp_units, rl_in + rl_out
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--workload", choices=["proof", "msm", "ntt", "verify"], default="proof")
+    ap.add_argument("--workload",
+                    choices=["ptx", "proof", "msm", "ntt", "verify"],
+                    default="ptx")
     ap.add_argument("--streams", type=int, default=0,
                     help="concurrent proving contexts per GPU (proof workload); "
                          "0 = auto, scaled to this rank's CPU share")
@@ -97,43 +165,75 @@ def main():
     gpu = taiga_amd.TaigaGpu(local_rank)
 
     # ---- setup (untimed): synthetic inputs resident in HBM ----
-    if args.workload == "proof":
-        # BASELINE configs[3/4] class: Action-circuit-shaped proofs (k=15).
-        # PK + SRS resident on the GPU; witness/instance derived from seeds
-        # per step (fresh randomness each step — nothing cached). With
-        # --streams C > 1, C independent proving contexts on the SAME device
-        # overlap one proof's host phases with another's kernels (ctypes
-        # releases the GIL during C calls); a step = C proofs.
+    if args.workload in ("ptx", "proof"):
+        # THE REAL CIRCUITS (round 2): the exact Compliance (Action) and
+        # TrivialRL constraint systems, restated from
+        # compliance_circuit.rs / resource_logic_examples.rs
+        # (tools/circuit; GPU proof bytes bit-identical to the CPU oracle,
+        # tests/test_compliance_circuit.py).
+        #   ptx   (default) = BASELINE configs[3]: one step = one full
+        #           ShieldedPartialTransaction build per context — 2
+        #           compliance + 4 TrivialRL proofs over the ptx resource
+        #           tree, assembled into the borsh bundle
+        #           (shielded_ptx.rs:98-137). value counts ACTION
+        #           (compliance) proofs: 2 per ptx.
+        #   proof = one compliance proof per context per step.
+        # Witness units (borsh ComplianceInfo / ResourceExistenceWitness)
+        # are caller inputs, generated once per rank (the reference's
+        # criterion bench likewise re-proves one fixed ComplianceInfo —
+        # benches/compliance_proof.rs:84-96); witness SYNTHESIS, proving
+        # and bundling run fully inside the timed step, with a fresh rng
+        # seed per step. With --streams C > 1, C independent contexts on
+        # the same device overlap host phases with kernels; a step = C
+        # units.
         import concurrent.futures
         import pathlib
 
         golden = pathlib.Path(REPO) / "tests" / "golden"
         srs_bytes = (golden / "params_15").read_bytes()
-        desc_bytes = (golden / "cs1.desc").read_bytes()
-        # auto stream count: 8 concurrent contexts when this rank has a
-        # full CPU share (single rank, or an 8-GPU node with a big cgroup
-        # quota); fall back when ranks split a small quota (host witness/
-        # transcript stages would oversubscribe and stall)
+        cdesc = (golden / "compliance.desc").read_bytes()
+        ctgw = (golden / "compliance.tgw").read_bytes()
+        rdesc = (golden / "trivial_rl.desc").read_bytes()
+        rtgw = (golden / "trivial_rl.tgw").read_bytes()
         C = args.streams if args.streams > 0 else (
             8 if N_CORES >= 12 else (4 if N_CORES >= 6 else 2))
         args.streams = C
         ctxs = [gpu] + [taiga_amd.TaigaGpu(local_rank) for _ in range(C - 1)]
+        slot_c = slot_r = 0
         for g in ctxs:
             g.load_srs(srs_bytes)
-            g.keygen(desc_bytes)
+            slot_c = g.keygen(cdesc)
+            g.witness_program_load(ctgw)
+            if args.workload == "ptx":
+                slot_r = g.keygen(rdesc)
+                g.witness_program_load(rtgw)
+            g.select_key(slot_c)
         pool = concurrent.futures.ThreadPoolExecutor(max_workers=C)
         counter = [0]
+        comp_units, rl_units = _make_ptx_units(rank)
+        import ctypes as _ct
+        lib = taiga_amd.api.load_library()
 
-        def one_proof(g, i):
-            inst = (SEED + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
-            wit = (SEED + 77 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
-            rng_s = (SEED + 99 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
-            g.create_proof(inst, wit, rng_s)
+        if args.workload == "ptx":
+            bufs = [_ct.create_string_buffer(1 << 18) for _ in range(C)]
+
+            def one_unit(j, i):
+                rng_s = (SEED + 99 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+                out_len = _ct.c_size_t()
+                rc = lib.tg_ptx_build(ctxs[j]._h, slot_c, slot_r, 2, comp_units,
+                                      2, 2, rl_units, rng_s, bufs[j],
+                                      len(bufs[j]), _ct.byref(out_len))
+                assert rc == 0, f"tg_ptx_build rc={rc}"
+        else:
+
+            def one_unit(j, i):
+                rng_s = (SEED + 99 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
+                ctxs[j].compliance_prove(comp_units[:1528], rng_s)
 
         def step():
             base = counter[0]
             counter[0] += C
-            futs = [pool.submit(one_proof, ctxs[j], base + j) for j in range(C)]
+            futs = [pool.submit(one_unit, j, base + j) for j in range(C)]
             for f in futs:
                 f.result()
     elif args.workload == "verify":
@@ -146,17 +246,18 @@ def main():
 
         golden = pathlib.Path(REPO) / "tests" / "golden"
         gpu.load_srs((golden / "params_15").read_bytes())
-        gpu.keygen((golden / "cs1.desc").read_bytes())
+        gpu.keygen((golden / "compliance.desc").read_bytes())
+        gpu.witness_program_load((golden / "compliance.tgw").read_bytes())
+        comp_units, _ = _make_ptx_units(rank)
         BUNDLE = 6
         items = []
         for i in range(BUNDLE):
-            inst = (SEED + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
-            wit = (SEED + 77 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
             rng_s = (SEED + 99 + rank).to_bytes(16, "little") + i.to_bytes(16, "little")
-            items.append((inst, gpu.create_proof(inst, wit, rng_s)))
+            proof, inst = gpu.compliance_prove(comp_units[(i % 2) * 1528:(i % 2 + 1) * 1528], rng_s)
+            items.append((inst, proof))
 
         def step():
-            assert gpu.verify_batch(items)
+            assert gpu.verify_batch_raw(items)
     elif args.workload == "msm":
         gpu.gen_bases(MSM_N, SEED)  # same base set on every rank
         scalars = gen_scalars(MSM_N, SEED + 1000 + rank)  # per-rank scalars
@@ -203,22 +304,44 @@ def main():
         elapsed = float(t.item())
 
     # ---- roofline (dominant kernel, HIP events on the launch stream) ----
-    if args.workload == "proof":
+    # For the concurrent proof workloads, the timed region's per-launch
+    # averages are stretched by C contexts time-sharing the chip; re-run
+    # ONE unit uncontended with fresh counters so the roofline reflects
+    # kernel quality, not contention (VERDICT round-1 item).
+    if args.workload in ("ptx", "proof") and rank == 0:
+        gpu.prof_reset()
+        one_unit(0, 10_000_000)
+        gpu.synchronize()
+    if args.workload in ("ptx", "proof"):
         acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
         # aggregate per-step algorithmic bytes over the dominant kernel's
-        # launches (batched, so per-launch sizes differ): one proof gathers
-        # ~29n points of 16-window bucket work (commits: 27 column-size MSMs;
-        # IPA rounds halve geometrically to ~2n) x (64 B base + 4 B index).
-        # Reported per average launch for the contract's avg-launch framing.
-        alg_bytes_per_proof = 16 * 29 * (1 << 15) * 68
-        alg_bytes = (alg_bytes_per_proof * args.steps / acc_n) if acc_n else 0
+        # launches (batched, so per-launch sizes differ): one compliance
+        # proof gathers ~29n points of 13-window bucket work (27
+        # column-size MSM commits; IPA rounds halve geometrically to ~2n)
+        # x (64 B base + 4 B index); a ptx holds 2 compliance + 4 RL
+        # proofs of the same k. Reported per average launch.
+        win = 13  # size-adaptive window at n=2^15
+        proofs_per_unit = 6 if args.workload == "ptx" else 1
+        alg_bytes_per_proof = win * 29 * (1 << 15) * 68
+        # counters hold the single uncontended probe unit (see above)
+        alg_bytes = (alg_bytes_per_proof * proofs_per_unit / acc_n) if acc_n else 0
         dom = ("msm_bucket_acc", acc_ms, acc_n, alg_bytes)
-        units_per_step = max(1, args.streams)
+        # value counts ACTION (compliance) proofs — BASELINE's metric
+        units_per_step = (2 if args.workload == "ptx" else 1) * max(1, args.streams)
         unit = "proofs/s"
         metric = "action_proofs_per_sec"
-        workload_name = ("compliance_shaped_proof_k15 (CS1 stand-in: same shape/size as the "
-                         "Action circuit — 10 advice, lookup, 12-col permutation, degree 9; "
-                         "exact compliance witness fidelity is the round-2 item)")
+        if args.workload == "ptx":
+            workload_name = (
+                "shielded_ptx_build (BASELINE configs[3]: one full "
+                "ShieldedPartialTransaction = 2 EXACT Compliance/Action proofs "
+                "+ 4 TrivialRL proofs, k=15, borsh bundle out; GPU proof bytes "
+                "bit-identical to the CPU oracle — tests/test_ptx.py)")
+        else:
+            workload_name = (
+                "compliance_proof_k15 (the EXACT Action circuit — "
+                "compliance_circuit.rs restated: Poseidon/ECC/Blake2s/Merkle "
+                "witness chain, degree 17, ext 2^19; bit-identical to the "
+                "oracle — tests/test_compliance_circuit.py)")
     elif args.workload == "verify":
         acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
         # one combined g-MSM (n=2^15) per batch regardless of bundle size
@@ -228,7 +351,7 @@ def main():
         unit = "proofs/s"
         metric = "action_proofs_verified_per_sec"
         workload_name = ("batch_verify_bundle6_k15 (one combined IPA check per "
-                         "6-proof bundle; CS1 Action-shaped circuit)")
+                         "6-proof bundle; EXACT compliance circuit)")
     elif args.workload == "msm":
         acc_ms, acc_n = gpu.prof_get("msm_bucket_acc")
         # algorithmic bytes per k_bucket_acc launch (BASELINE.md config 2
@@ -288,43 +411,89 @@ def main():
         import oracle_ct as oc
 
         cores = N_CORES
-        if args.workload == "proof":
-            lib = oc.lib()
-            import pathlib
-
-            golden = pathlib.Path(REPO) / "tests" / "golden"
-            desc = (golden / "cs1.desc").read_bytes()
-            srs = (golden / "params_15").read_bytes()
-            assert lib.orc_prover_init(desc, len(desc), srs, len(srs)) in (0, 1)
+        if args.workload in ("ptx", "proof"):
+            # oracle (the CPU restatement, "port") on the SAME real
+            # circuits; bounded sample = one compliance proof (+ one RL
+            # proof for the ptx rate), keygen excluded (the product
+            # caches PKs too). ptx rate model: t_ptx = 2*t_c + 4*t_rl.
             import ctypes
-
-            lib.orc_prove_cs1.restype = ctypes.c_long
-            out = ctypes.create_string_buffer(1 << 14)
-            inst = (SEED).to_bytes(32, "little")
-            wit = (SEED + 77).to_bytes(32, "little")
-            rng_s = (SEED + 99).to_bytes(32, "little")
-            t0 = time.perf_counter()
-            nlen = lib.orc_prove_cs1(inst, wit, rng_s, out, 1 << 14)
-            dt = time.perf_counter() - t0
-            assert nlen > 0
-            cpu_baseline = {
-                "value": round(1.0 / dt, 4),
-                "unit": unit,
-                "cores": cores,
-                "kind": "port",
-                "sample": "one CS1 proof via the oracle prover (OpenMP where parallel)",
-            }
-        elif args.workload == "verify":
-            lib = oc.lib()
             import pathlib
 
+            lib = oc.lib()
+            lib.orc_prove_raw.restype = ctypes.c_long
             golden = pathlib.Path(REPO) / "tests" / "golden"
-            desc = (golden / "cs1.desc").read_bytes()
             srs = (golden / "params_15").read_bytes()
-            assert lib.orc_prover_init(desc, len(desc), srs, len(srs)) in (0, 1)
+
+            def orc_time_one(name, kind):
+                desc = (golden / f"{name}.desc").read_bytes()
+                tgw = (golden / f"{name}.tgw").read_bytes()
+                lib.orc_prover_reset()
+                assert lib.orc_prover_init(desc, len(desc), srs, len(srs)) == 0
+                prog = ctypes.c_void_p()
+                assert lib.orc_tgw_load(tgw, ctypes.c_long(len(tgw)),
+                                        ctypes.byref(prog)) == 0
+                n = 1 << 15
+                adv = ctypes.create_string_buffer(10 * n * 32)
+                if kind == 0:
+                    inputs = ctypes.create_string_buffer(124 * 32)
+                    assert lib.orc_compliance_inputs(
+                        comp_units[:1528], ctypes.c_long(1528), inputs) == 0
+                    ninst = 9
+                    inst = bytearray(ninst * 32)
+                    inst[32:64] = comp_units[1262:1294]  # anchor (202+4+33*32)
+                else:
+                    inputs = ctypes.create_string_buffer(41 * 32)
+                    padding = ctypes.create_string_buffer(16 * 32)
+                    assert lib.orc_rl_inputs(rl_units[:334], ctypes.c_long(334),
+                                             bytes(32), inputs, padding) == 0
+                    ninst = 22
+                    inst = bytearray(ninst * 32)
+                    inst[6 * 32:] = padding.raw
+                t0 = time.perf_counter()
+                assert lib.orc_tgw_run(prog, inputs, 10, adv) == 0
+                buf = (ctypes.c_char * len(inst)).from_buffer(inst)
+                assert lib.orc_tgw_instance(prog, 10, adv, buf) == 0
+                out = ctypes.create_string_buffer(1 << 14)
+                plen = lib.orc_prove_raw(bytes(inst), adv, bytes(32), out,
+                                         ctypes.c_long(1 << 14))
+                dt = time.perf_counter() - t0
+                lib.orc_tgw_free(prog)
+                assert plen > 0, plen
+                return dt
+
+            t_c = orc_time_one("compliance", 0)
+            if args.workload == "ptx":
+                t_r = orc_time_one("trivial_rl", 1)
+                cpu_baseline = {
+                    "value": round(2.0 / (2 * t_c + 4 * t_r), 4),
+                    "unit": unit,
+                    "cores": cores,
+                    "kind": "port",
+                    "sample": ("one exact-compliance prove (%.1fs) + one "
+                               "TrivialRL prove (%.1fs) on the oracle; ptx "
+                               "rate = 2 action proofs / (2*t_c + 4*t_rl), "
+                               "keygen excluded" % (t_c, t_r)),
+                }
+            else:
+                cpu_baseline = {
+                    "value": round(1.0 / t_c, 4),
+                    "unit": unit,
+                    "cores": cores,
+                    "kind": "port",
+                    "sample": "one exact-compliance proof via the oracle (witness synthesis + prove)",
+                }
+        elif args.workload == "verify":
+            import pathlib
+
+            lib = oc.lib()
+            golden = pathlib.Path(REPO) / "tests" / "golden"
+            desc = (golden / "compliance.desc").read_bytes()
+            srs = (golden / "params_15").read_bytes()
+            lib.orc_prover_reset()
+            assert lib.orc_prover_init(desc, len(desc), srs, len(srs)) == 0
             inst, proof = items[0]
             t0 = time.perf_counter()
-            assert lib.orc_verify_cs1(inst, proof, len(proof)) == 0
+            assert lib.orc_verify_raw(inst, proof, len(proof)) == 0
             dt = time.perf_counter() - t0
             cpu_baseline = {
                 "value": round(1.0 / dt, 4),
@@ -378,8 +547,8 @@ def main():
             "data": "synthetic",
             "config": {
                 "workload": workload_name,
-                "n_points": {"proof": 1 << 15, "verify": 1 << 15, "msm": MSM_N, "ntt": 1 << NTT_K}[args.workload],
-                "window_bits": 16 if args.workload in ("proof", "verify", "msm") else None,
+                "n_points": {"ptx": 1 << 15, "proof": 1 << 15, "verify": 1 << 15, "msm": MSM_N, "ntt": 1 << NTT_K}[args.workload],
+                "window_bits": 16 if args.workload == "msm" else (13 if args.workload in ("ptx", "proof", "verify") else None),
                 "parallelism": f"dp{n_gpus} (independent proofs/MSMs per GPU, no collective — SURVEY §8e)",
             },
             "roofline": roofline,
