@@ -406,7 +406,7 @@ def main():
 
     # ---- CPU baseline (oracle "port", rank 0, N=1 only) ----
     cpu_baseline = None
-    if rank == 0 and world <= 1:
+    if rank == 0 and world <= 1 and not os.environ.get("TG_SKIP_CPU_BASELINE"):
         sys.path.insert(0, os.path.join(REPO, "oracle"))
         import oracle_ct as oc
 

@@ -213,12 +213,32 @@ int tg_tx_digest(const uint8_t* nfs, size_t n_nf, const uint8_t* cms, size_t n_c
  * vk_len = byte length of one embedded RL VerifyingKey (32*(n_fixed +
  * n_perm) for TGD1 keys). TG_OK; -1 bad binding signature; -2xx
  * structural decode error.
- * tg_tx_verify: the above with vk_len from the ACTIVE key, plus ONE
- * combined batch verification of every compliance proof in the bundle
- * (requires the active circuit's n_instance_rows <= 6). */
+ * tg_tx_verify: LEGACY/PARTIAL (round 1): verifies compliance proofs +
+ * the binding signature only, with the round-1 first-n-rows instance
+ * mapping (requires n_instance_rows <= 6); resource-logic proofs are
+ * parsed but NOT verified — TG_OK from this entry does NOT mean the
+ * transaction is fully valid.
+ * tg_tx_verify_full: the COMPLETE check (round 2): Transaction::execute
+ * semantics — wire parse + Transaction::digest + binding signature +
+ * batch verification of EVERY compliance proof (real 9-row instance
+ * expansion: delta decompressed, RL-commitment halves split —
+ * compliance.rs to_instance) and EVERY resource-logic proof (vk bytes
+ * must match slot_rl's key), plus the per-sptx nf/cm/self-id/root
+ * consistency checks. */
 int tg_tx_wire_check(const uint8_t* tx, size_t len, uint32_t vk_len,
                      uint32_t* n_sptx, uint32_t* n_proofs);
 int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len);
+int tg_tx_verify_full(tg_ctx* ctx, int slot_compliance, int slot_rl,
+                      const uint8_t* tx, size_t len);
+/* ShieldedPartialTransaction build/verify (shielded_ptx.rs:98-137;
+ * BASELINE configs[3]): see taiga_gpu.cpp headers for unit layouts. */
+int tg_ptx_build(tg_ctx* ctx, int slot_compliance, int slot_rl,
+                 uint32_t n_compliance, const uint8_t* compliance_units,
+                 uint32_t n_in, uint32_t n_out, const uint8_t* rl_units,
+                 const uint8_t rng_seed[32], uint8_t* ptx_out, size_t cap,
+                 size_t* out_len);
+int tg_ptx_verify(tg_ctx* ctx, int slot_compliance, int slot_rl,
+                  const uint8_t* ptx, size_t len);
 
 /* ---- kernel profiling (HIP events on the ctx stream) ----
  * names: "msm_digits", "msm_scan", "msm_scatter", "msm_bucket_acc",

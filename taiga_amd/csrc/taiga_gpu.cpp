@@ -929,6 +929,90 @@ int tg_tx_verify(tg_ctx* ctx, const uint8_t* tx, size_t len) {
   return pverify_eval(c, *c->ppk, gds.data(), (int)m, rho.data());
 }
 
+static bool compliance_rows_expand(const uint8_t* inst192, uint8_t rows[288]);
+static std::vector<uint8_t> pk_vk_bytes(PPk* k);
+
+/* FULL transaction verification (round 2; supersedes tg_tx_verify's
+ * compliance-only check — ADVICE.md item 2): borsh Transaction parse +
+ * Transaction::digest + binding signature (tx_check), then batch-verify
+ * EVERY compliance proof (slot_compliance; 192-B instances expanded to
+ * the real 9 rows) and EVERY resource-logic proof (slot_rl; vk bytes
+ * must match the slot), plus the per-sptx execute() consistency checks.
+ * TG_OK only when the whole transaction is valid. */
+int tg_tx_verify_full(tg_ctx* ctx, int slot_compliance, int slot_rl,
+                      const uint8_t* tx, size_t len) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  int rc = tg_select_key(ctx, slot_rl);
+  if (rc) return rc;
+  PPk* rlk = c->ppk;
+  uint32_t vk_len = 32u * (uint32_t)(rlk->fixed_commits.size() +
+                                     rlk->sigma_commits.size());
+  TxDigestStreams st;
+  rc = tx_check(tx, len, vk_len, nullptr, nullptr, &st);
+  if (rc) return rc;
+  // vk match + consistency checks per sptx
+  std::vector<uint8_t> vk_ref = pk_vk_bytes(rlk);
+  for (const uint8_t* vkb : st.rl_vk_ptr)
+    if (memcmp(vkb, vk_ref.data(), vk_len) != 0) return -207;
+  st.sptx_comp_begin.push_back(st.n_compliance);
+  st.sptx_rl_begin.push_back(st.n_rl);
+  std::vector<uint8_t> cinst;
+  for (uint32_t i = 0; i < st.n_compliance; i++) {
+    uint8_t rows[288];
+    if (!compliance_rows_expand(st.inst_ptr[i], rows)) return -204;
+    cinst.insert(cinst.end(), rows, rows + 288);
+  }
+  for (uint32_t s2 = 0; s2 < st.n_sptx; s2++) {
+    uint32_t c0 = st.sptx_comp_begin[s2], c1 = st.sptx_comp_begin[s2 + 1];
+    uint32_t r0 = st.sptx_rl_begin[s2], r1 = st.sptx_rl_begin[s2 + 1];
+    if (r1 == r0) continue;
+    const uint8_t* root0 = st.rl_inst_ptr[r0];
+    uint32_t in_seen = 0, out_seen = 0;
+    for (uint32_t r = r0; r < r1; r++) {
+      if (memcmp(st.rl_inst_ptr[r], root0, 32) != 0) return -301;
+      const uint8_t* self_id = st.rl_inst_ptr[r] + 32;
+      if (st.rl_is_input[r]) {
+        uint32_t ci = c0 + in_seen++;
+        if (ci >= c1 || memcmp(self_id, cinst.data() + 288ul * ci, 32) != 0)
+          return -303;
+      } else {
+        uint32_t ci = c0 + out_seen++;
+        if (ci >= c1 || memcmp(self_id, cinst.data() + 288ul * ci + 64, 32) != 0)
+          return -304;
+      }
+    }
+  }
+  // batch-verify RL proofs (RL key active)
+  if (st.n_rl) {
+    std::vector<uint8_t> insts;
+    std::vector<uint8_t> proofs;
+    std::vector<size_t> lens;
+    for (uint32_t r = 0; r < st.n_rl; r++) {
+      insts.insert(insts.end(), st.rl_inst_ptr[r], st.rl_inst_ptr[r] + 22 * 32);
+      proofs.insert(proofs.end(), st.rl_proof_ptr[r],
+                    st.rl_proof_ptr[r] + st.rl_proof_len[r]);
+      lens.push_back(st.rl_proof_len[r]);
+    }
+    rc = tg_verify_batch_raw(ctx, st.n_rl, insts.data(), proofs.data(),
+                             lens.data());
+    if (rc) return rc == TG_ERR_BADARG ? rc : -1;
+  }
+  // batch-verify compliance proofs
+  rc = tg_select_key(ctx, slot_compliance);
+  if (rc) return rc;
+  if (st.n_compliance) {
+    std::vector<uint8_t> proofs;
+    for (uint32_t i = 0; i < st.n_compliance; i++)
+      proofs.insert(proofs.end(), st.proof_ptr[i],
+                    st.proof_ptr[i] + st.proof_len[i]);
+    rc = tg_verify_batch_raw(ctx, st.n_compliance, cinst.data(), proofs.data(),
+                             st.proof_len.data());
+    if (rc) return rc == TG_ERR_BADARG ? rc : -1;
+  }
+  return TG_OK;
+}
+
 /* batched Poseidon P128Pow5T3 ConstantLength<L> hashing (GPU witness
  * synthesis; replaces host-side halo2_gadgets poseidon
  * hashing — utils.rs:40-48): msgs = n x L x 32B canonical reprs,
@@ -1208,6 +1292,49 @@ int tg_ptx_build(tg_ctx* ctx, int slot_compliance, int slot_rl,
   return TG_OK;
 }
 
+/* expand the 192-byte borsh CompliancePublicInputs block into the real
+ * 9 instance rows [nf, anchor, cm, delta_x, delta_y, rlcm halves]
+ * (compliance.rs to_instance order; closes the round-1 placeholder
+ * mapping flagged in ADVICE.md item 3) */
+static bool compliance_rows_expand(const uint8_t* inst192, uint8_t rows[288]) {
+  memcpy(rows, inst192 + 32, 32);       // nf
+  memcpy(rows + 32, inst192, 32);       // anchor
+  memcpy(rows + 64, inst192 + 64, 32);  // cm
+  bool all0 = true;
+  for (int b = 0; b < 32; b++)
+    if (inst192[96 + b]) all0 = false;
+  if (all0) {
+    memset(rows + 96, 0, 64);
+  } else {
+    PallasJac d;
+    if (!pallas_decompress(d, inst192 + 96)) return false;
+    PallasAff a = jac_to_aff(d);
+    Fp x = fd_from_mont(a.x), y = fd_from_mont(a.y);
+    memcpy(rows + 96, x.l, 32);
+    memcpy(rows + 128, y.l, 32);
+  }
+  for (int half = 0; half < 2; half++) {
+    memset(rows + 160 + 64 * half, 0, 64);
+    memcpy(rows + 160 + 64 * half, inst192 + 128 + 32 * half, 16);
+    memcpy(rows + 192 + 64 * half + 0, inst192 + 128 + 32 * half + 16, 16);
+  }
+  return true;
+}
+
+static std::vector<uint8_t> pk_vk_bytes(PPk* k) {
+  std::vector<uint8_t> out;
+  uint8_t b[32];
+  for (const VestaAff& p : k->fixed_commits) {
+    Transcript::compress(b, p);
+    out.insert(out.end(), b, b + 32);
+  }
+  for (const VestaAff& p : k->sigma_commits) {
+    Transcript::compress(b, p);
+    out.insert(out.end(), b, b + 32);
+  }
+  return out;
+}
+
 /* verify one borsh ShieldedPartialTransaction end-to-end
  * (ShieldedPartialTransaction::execute, shielded_ptx.rs:232-240):
  * batch-verify ALL compliance proofs (slot_compliance) and ALL RL proofs
@@ -1239,32 +1366,8 @@ int tg_ptx_verify(tg_ctx* ctx, int slot_compliance, int slot_rl,
     cplen.push_back(plen);
     const uint8_t* inst;
     if (!cur.take(inst, 192)) return -203;
-    // expand the 192B borsh block into the 9 instance rows
-    // [nf, anchor, cm, dx, dy, in1, in2, out1, out2]
     uint8_t rows[288];
-    memcpy(rows, inst + 32, 32);       // nf
-    memcpy(rows + 32, inst, 32);       // anchor
-    memcpy(rows + 64, inst + 64, 32);  // cm
-    {
-      PallasJac d;
-      bool all0 = true;
-      for (int b = 0; b < 32; b++)
-        if (inst[96 + b]) all0 = false;
-      if (all0) {
-        memset(rows + 96, 0, 64);
-      } else {
-        if (!pallas_decompress(d, inst + 96)) return -204;
-        PallasAff a = jac_to_aff(d);
-        Fp x = fd_from_mont(a.x), y = fd_from_mont(a.y);
-        memcpy(rows + 96, x.l, 32);
-        memcpy(rows + 128, y.l, 32);
-      }
-    }
-    for (int half = 0; half < 2; half++) {
-      memset(rows + 160 + 64 * half, 0, 64);
-      memcpy(rows + 160 + 64 * half, inst + 128 + 32 * half, 16);
-      memcpy(rows + 192 + 64 * half, inst + 128 + 32 * half + 16, 16);
-    }
+    if (!compliance_rows_expand(inst, rows)) return -204;
     cinst.insert(cinst.end(), rows, rows + 288);
     nfs.insert(nfs.end(), rows, rows + 32);
     cms.insert(cms.end(), rows + 64, rows + 96);
@@ -1281,21 +1384,9 @@ int tg_ptx_verify(tg_ctx* ctx, int slot_compliance, int slot_rl,
     for (uint32_t i = 0; i < cnt; i++) {
       const uint8_t* vkb;
       if (!cur.take(vkb, vk_len)) return -206;
-      if (memcmp(vkb, [&] {
-            static thread_local std::vector<uint8_t> vk_ref;
-            vk_ref.clear();
-            for (const VestaAff& p : rlk->fixed_commits) {
-              uint8_t b[32];
-              Transcript::compress(b, p);
-              vk_ref.insert(vk_ref.end(), b, b + 32);
-            }
-            for (const VestaAff& p : rlk->sigma_commits) {
-              uint8_t b[32];
-              Transcript::compress(b, p);
-              vk_ref.insert(vk_ref.end(), b, b + 32);
-            }
-            return vk_ref.data();
-          }(), vk_len) != 0)
+      static thread_local std::vector<uint8_t> vk_ref;
+      if (vk_ref.size() != vk_len) vk_ref = pk_vk_bytes(rlk);
+      if (memcmp(vkb, vk_ref.data(), vk_len) != 0)
         return -207;  // unknown RL vk (only the TrivialRL key is loaded)
       uint32_t plen;
       const uint8_t* p;

@@ -72,25 +72,43 @@ struct TxDigestStreams {
   std::vector<size_t> proof_len;
   std::vector<const uint8_t*> inst_ptr;  // 192B each
   const uint8_t* sig = nullptr;          // 64B binding signature
+  // resource-logic proof views (round 2: RL proofs are verified too)
+  std::vector<const uint8_t*> rl_vk_ptr, rl_proof_ptr, rl_inst_ptr;  // inst 22x32
+  std::vector<size_t> rl_proof_len;
+  std::vector<uint8_t> rl_is_input;  // 1 = inputs group
+  // per-sptx boundaries: [compliance_begin, rl_begin] index pairs
+  std::vector<uint32_t> sptx_comp_begin, sptx_rl_begin;
 };
 
-// one ResourceLogicVerifyingInfo (structure only)
-inline bool tx_parse_rl_info(TxCursor& c, uint32_t vk_len) {
-  const uint8_t* b;
-  if (!c.take(b, vk_len)) return false;
+// one ResourceLogicVerifyingInfo; records views when st != nullptr
+inline bool tx_parse_rl_info(TxCursor& c, uint32_t vk_len, TxDigestStreams* st,
+                             uint8_t is_input) {
+  const uint8_t* vk;
+  if (!c.take(vk, vk_len)) return false;
   uint32_t plen;
   if (!c.u32(plen) || plen > (1u << 20)) return false;
-  if (!c.take(b, plen)) return false;
-  return c.take(b, 32 * TX_RL_PUBLIC_INPUTS);
+  const uint8_t* proof;
+  if (!c.take(proof, plen)) return false;
+  const uint8_t* inst;
+  if (!c.take(inst, 32 * TX_RL_PUBLIC_INPUTS)) return false;
+  if (st) {
+    st->rl_vk_ptr.push_back(vk);
+    st->rl_proof_ptr.push_back(proof);
+    st->rl_proof_len.push_back(plen);
+    st->rl_inst_ptr.push_back(inst);
+    st->rl_is_input.push_back(is_input);
+  }
+  return true;
 }
 
-inline bool tx_parse_rl_set(TxCursor& c, uint32_t vk_len, uint32_t& n_rl) {
-  if (!tx_parse_rl_info(c, vk_len)) return false;
+inline bool tx_parse_rl_set(TxCursor& c, uint32_t vk_len, uint32_t& n_rl,
+                            TxDigestStreams* st, uint8_t is_input) {
+  if (!tx_parse_rl_info(c, vk_len, st, is_input)) return false;
   n_rl++;
   uint32_t n;
   if (!c.u32(n) || n > TX_MAX_ITEMS) return false;
   for (uint32_t i = 0; i < n; i++) {
-    if (!tx_parse_rl_info(c, vk_len)) return false;
+    if (!tx_parse_rl_info(c, vk_len, st, is_input)) return false;
     n_rl++;
   }
   return true;
@@ -105,6 +123,8 @@ inline int tx_parse(const uint8_t* tx, size_t len, uint32_t vk_len,
   if (!c.u32(n_sptx) || n_sptx > TX_MAX_ITEMS) return -201;
   out.n_sptx = n_sptx;
   for (uint32_t s = 0; s < n_sptx; s++) {
+    out.sptx_comp_begin.push_back(out.n_compliance);
+    out.sptx_rl_begin.push_back(out.n_rl);
     uint32_t n_cvi;
     if (!c.u32(n_cvi) || n_cvi > TX_MAX_ITEMS) return -202;
     for (uint32_t i = 0; i < n_cvi; i++) {
@@ -127,10 +147,10 @@ inline int tx_parse(const uint8_t* tx, size_t len, uint32_t vk_len,
     uint32_t n_in, n_out;
     if (!c.u32(n_in) || n_in > TX_MAX_ITEMS) return -206;
     for (uint32_t i = 0; i < n_in; i++)
-      if (!tx_parse_rl_set(c, vk_len, out.n_rl)) return -207;
+      if (!tx_parse_rl_set(c, vk_len, out.n_rl, &out, 1)) return -207;
     if (!c.u32(n_out) || n_out > TX_MAX_ITEMS) return -208;
     for (uint32_t i = 0; i < n_out; i++)
-      if (!tx_parse_rl_set(c, vk_len, out.n_rl)) return -209;
+      if (!tx_parse_rl_set(c, vk_len, out.n_rl, &out, 0)) return -209;
     const uint8_t* b;
     if (!c.take(b, 1)) return -210;
     if (*b == 1) {
