@@ -264,13 +264,18 @@ __global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, co
   }
 }
 
-// phase 2: one 64-lane wave per big bucket; lane-strided partials + LDS tree
+// phase 2: one 256-lane block per big bucket; lane-strided partials + LDS
+// tree. 256 lanes (vs round-1's 64): the real circuits' bit/byte-valued
+// advice columns (blake2s decompositions) concentrate ~n/2 points into
+// single buckets — the wide block cuts the serial run per lane 4x
+// (k_bucket_acc_big was 15.8% of proof GPU time).
+constexpr int MSM_BIG_LANES = 256;
 template <bool SAFE>
-__global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
+__global__ void __launch_bounds__(MSM_BIG_LANES, 1) k_bucket_acc_big(const uint32_t* start,
                                  const uint32_t* end, const uint32_t* sorted,
                                  const VestaAff* pts, VestaJac* buckets,
                                  const uint32_t* big_list, const uint32_t* big_count) {
-  __shared__ VestaJac lds[64];
+  __shared__ VestaJac lds[MSM_BIG_LANES];
   uint32_t nbig = *big_count;
   for (uint32_t gi = blockIdx.x; gi < nbig; gi += gridDim.x) {
     u64 b = big_list[gi];
@@ -286,7 +291,7 @@ __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
         acc.x = p0.x;
         acc.y = p0.y;
         acc.z = fd_one_mont<FqCfg>();
-        for (uint32_t idx = first + 64; idx < e; idx += 64) {
+        for (uint32_t idx = first + MSM_BIG_LANES; idx < e; idx += MSM_BIG_LANES) {
           uint32_t ent = sorted[idx];
           VestaAff p = pts[ent & 0x7FFFFFFFu];
           if (ent >> 31) p = aff_neg_fast(p);
@@ -294,7 +299,7 @@ __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
         }
       }
     } else {
-      for (uint32_t idx = s + t; idx < e; idx += 64) {
+      for (uint32_t idx = s + t; idx < e; idx += MSM_BIG_LANES) {
         uint32_t ent = sorted[idx];
         VestaAff p = pts[ent & 0x7FFFFFFFu];
         if (ent >> 31) p = aff_neg(p);
@@ -303,7 +308,7 @@ __global__ void __launch_bounds__(64, 1) k_bucket_acc_big(const uint32_t* start,
     }
     lds[t] = acc;
     __syncthreads();
-    for (int off = 32; off >= 1; off >>= 1) {
+    for (int off = MSM_BIG_LANES / 2; off >= 1; off >>= 1) {
       if (t < off) lds[t] = jac_add(lds[t], lds[t + off]);
       __syncthreads();
     }

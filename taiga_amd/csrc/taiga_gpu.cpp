@@ -22,6 +22,7 @@
 #include "fd28.hpp"
 #include "witness.hpp"
 #include "prover_impl.hpp"
+#include "hf_rtc.hpp"
 
 namespace taiga {
 
@@ -497,14 +498,14 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
         hipLaunchKernelGGL(k_bucket_acc<true>, dim3(msm_grid(m)), dim3(256), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
                            c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m);
-        hipLaunchKernelGGL(k_bucket_acc_big<true>, dim3(1024), dim3(64), 0, c->stream,
+        hipLaunchKernelGGL(k_bucket_acc_big<true>, dim3(1024), dim3(MSM_BIG_LANES), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
                            c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
       } else {
         hipLaunchKernelGGL(k_bucket_acc<false>, dim3(msm_grid(m)), dim3(256), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
                            c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m);
-        hipLaunchKernelGGL(k_bucket_acc_big<false>, dim3(1024), dim3(64), 0, c->stream,
+        hipLaunchKernelGGL(k_bucket_acc_big<false>, dim3(1024), dim3(MSM_BIG_LANES), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
                            c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
       }
