@@ -24,6 +24,7 @@
 #include <cstdio>
 #include <map>
 #include <mutex>
+#include <array>
 #include <sstream>
 #include <string>
 #include <vector>
@@ -129,11 +130,15 @@ __device__ __forceinline__ Fp f_mul(Fp a, Fp b) {
   return o.str();
 }
 
-// generate the hf_gates kernel source for this desc's gate constraints
+// generate the hf_gates kernel source for this desc's gate constraints.
+// One __noinline__ __device__ function PER GATE (a single fused basic
+// block of ~5k inlined Montgomery muls sent the optimizer into a
+// half-hour register-allocation spiral; per-gate functions compile in
+// seconds and still keep each gate's queries register-resident with
+// full common-subexpression elimination inside the gate).
 inline std::string hf_rtc_source(const PDesc& d) {
   std::ostringstream o;
   o << hf_rtc_prelude();
-  // constants in Montgomery form, as used by the interpreter
   o << "__device__ __constant__ Fp CONSTS[" << (d.consts.empty() ? 1 : d.consts.size())
     << "] = {";
   for (size_t i = 0; i < d.consts.size(); i++) {
@@ -143,94 +148,97 @@ inline std::string hf_rtc_source(const PDesc& d) {
   }
   if (d.consts.empty()) o << "{{0,0,0,0}}";
   o << "};\n";
-  o << "extern \"C\" __global__ void __launch_bounds__(256) hf_gates("
-       "Fp* out, Fp* const* cols, Fp y, long ext_n, long rs) {\n"
-       "  for (long i = blockIdx.x*(long)blockDim.x + threadIdx.x; i < ext_n;"
-       " i += (long)gridDim.x*blockDim.x) {\n";
-  // collect distinct queries used by gates
-  struct Q {
-    int kind, col, rot;
+
+  auto colbase = [&](uint32_t tag) {
+    return tag == XFIXED ? 0 : tag == XADVICE ? d.n_fixed : d.n_fixed + d.n_advice;
   };
-  std::vector<Q> qs;
-  auto qname = [&](int kind, int col, int rot) {
-    std::ostringstream n;
-    n << "q" << kind << "_" << col << "_" << (rot < 0 ? "m" : "p") << (rot < 0 ? -rot : rot);
-    return n.str();
-  };
-  auto qadd = [&](int kind, int col, int rot) {
-    for (const Q& q : qs)
-      if (q.kind == kind && q.col == col && q.rot == rot) return;
-    qs.push_back({kind, col, rot});
-  };
-  for (const auto& g : d.gates)
-    for (const auto& op : g.ops) {
-      if (op.tag == XFIXED) qadd(0, (int)op.a, op.b);
-      else if (op.tag == XADVICE) qadd(1, (int)op.a, op.b);
-      else if (op.tag == XINSTANCE) qadd(2, (int)op.a, op.b);
-    }
-  // cols layout: [fixed 0..n_fixed) [advice ..) [instance ..)
-  for (const Q& q : qs) {
-    int base = q.kind == 0 ? 0 : q.kind == 1 ? d.n_fixed : d.n_fixed + d.n_advice;
-    o << "    const Fp " << qname(q.kind, q.col, q.rot) << " = cols[" << (base + q.col)
-      << "][(i + (long)(" << q.rot << ")*rs) & (ext_n-1)];\n";
-  }
-  o << "    Fp acc = {{0,0,0,0}};\n";
-  int tmp = 0;
+
+  // per-gate function with local query loads + hash-consed temporaries
   for (size_t gi = 0; gi < d.gates.size(); gi++) {
-    // postfix -> straight-line temporaries
+    o << "__device__ __noinline__ Fp g" << gi
+      << "(Fp* const* cols, long i, long rs, long mask) {\n";
+    // distinct queries of this gate
+    std::vector<std::array<int, 3>> qs;
+    for (const auto& op : d.gates[gi].ops)
+      if (op.tag == XFIXED || op.tag == XADVICE || op.tag == XINSTANCE) {
+        std::array<int, 3> k{(int)op.tag, (int)op.a, op.b};
+        bool seen = false;
+        for (auto& q : qs)
+          if (q == k) { seen = true; break; }
+        if (!seen) qs.push_back(k);
+      }
+    auto qname = [&](uint32_t tag, uint32_t col, int rot) {
+      std::ostringstream n;
+      n << "q" << tag << "_" << col << "_" << (rot < 0 ? "m" : "p")
+        << (rot < 0 ? -rot : rot);
+      return n.str();
+    };
+    for (auto& q : qs) {
+      o << "  const Fp " << qname(q[0], q[1], q[2]) << " = cols["
+        << (colbase(q[0]) + q[1]) << "][(i + (long)(" << q[2] << ")*rs) & mask];\n";
+    }
+    // postfix -> temporaries with hash-consing (per gate)
+    std::map<std::string, std::string> cse;
+    int tmp = 0;
     std::vector<std::string> stk;
+    auto emit = [&](const std::string& expr) {
+      auto it = cse.find(expr);
+      if (it != cse.end()) return it->second;
+      std::ostringstream tn;
+      tn << "t" << tmp++;
+      o << "  const Fp " << tn.str() << " = " << expr << ";\n";
+      cse[expr] = tn.str();
+      return cse[expr];
+    };
     for (const auto& op : d.gates[gi].ops) {
-      char tn[16];
-      snprintf(tn, sizeof(tn), "t%d", tmp++);
-      std::string t = tn;
       switch (op.tag) {
-        case XCONST:
-          o << "    const Fp " << t << " = CONSTS[" << op.a << "];\n";
-          stk.push_back(t);
-          break;
-        case XFIXED: stk.push_back(qname(0, (int)op.a, op.b)); tmp--; break;
-        case XADVICE: stk.push_back(qname(1, (int)op.a, op.b)); tmp--; break;
-        case XINSTANCE: stk.push_back(qname(2, (int)op.a, op.b)); tmp--; break;
-        case XADD: {
-          std::string b = stk.back(); stk.pop_back();
-          std::string a = stk.back(); stk.pop_back();
-          o << "    const Fp " << t << " = f_add(" << a << "," << b << ");\n";
-          stk.push_back(t);
+        case XCONST: {
+          std::ostringstream e;
+          e << "CONSTS[" << op.a << "]";
+          stk.push_back(e.str());
           break;
         }
-        case XSUB: {
-          std::string b = stk.back(); stk.pop_back();
-          std::string a = stk.back(); stk.pop_back();
-          o << "    const Fp " << t << " = f_sub(" << a << "," << b << ");\n";
-          stk.push_back(t);
+        case XFIXED:
+        case XADVICE:
+        case XINSTANCE:
+          stk.push_back(qname(op.tag, op.a, op.b));
           break;
-        }
+        case XADD:
+        case XSUB:
         case XMUL: {
           std::string b = stk.back(); stk.pop_back();
           std::string a = stk.back(); stk.pop_back();
-          o << "    const Fp " << t << " = f_mul(" << a << "," << b << ");\n";
-          stk.push_back(t);
+          const char* fn = op.tag == XADD ? "f_add" : op.tag == XSUB ? "f_sub" : "f_mul";
+          stk.push_back(emit(std::string(fn) + "(" + a + "," + b + ")"));
           break;
         }
         case XNEG: {
           std::string a = stk.back(); stk.pop_back();
-          o << "    const Fp " << t << " = f_neg(" << a << ");\n";
-          stk.push_back(t);
+          stk.push_back(emit("f_neg(" + a + ")"));
           break;
         }
         case XSCALE: {
           std::string a = stk.back(); stk.pop_back();
-          o << "    const Fp " << t << " = f_mul(" << a << ", CONSTS[" << op.a
-            << "]);\n";
-          stk.push_back(t);
+          std::ostringstream e;
+          e << "f_mul(" << a << ",CONSTS[" << op.a << "])";
+          stk.push_back(emit(e.str()));
           break;
         }
         default:
-          return std::string();  // unknown op: no RTC
+          return std::string();
       }
     }
-    o << "    acc = f_add(f_mul(acc, y), " << stk.back() << ");\n";
+    o << "  return " << stk.back() << ";\n}\n";
   }
+
+  o << "extern \"C\" __global__ void __launch_bounds__(256) hf_gates("
+       "Fp* out, Fp* const* cols, Fp y, long ext_n, long rs) {\n"
+       "  long mask = ext_n - 1;\n"
+       "  for (long i = blockIdx.x*(long)blockDim.x + threadIdx.x; i < ext_n;"
+       " i += (long)gridDim.x*blockDim.x) {\n"
+       "    Fp acc = {{0,0,0,0}};\n";
+  for (size_t gi = 0; gi < d.gates.size(); gi++)
+    o << "    acc = f_add(f_mul(acc, y), g" << gi << "(cols, i, rs, mask));\n";
   o << "    out[i] = acc;\n  }\n}\n";
   return o.str();
 }
@@ -243,21 +251,25 @@ inline uint64_t hf_desc_hash(const PDesc& d) {
   return h;
 }
 
-inline HfRtcKernel* hf_rtc_get(const PDesc& d) {
-  static std::mutex mu;
-  static std::map<uint64_t, HfRtcKernel> cache;
-  if (getenv("TG_NO_RTC")) return nullptr;
-  std::lock_guard<std::mutex> lk(mu);
-  uint64_t h = hf_desc_hash(d);
-  auto it = cache.find(h);
-  if (it != cache.end()) return it->second.ready ? &it->second : nullptr;
-  HfRtcKernel& k = cache[h];  // default: not ready (negative-cache failures)
+inline std::string hf_cache_path(uint64_t h) {
+  const char* dir = getenv("TG_HF_CACHE");
+  if (!dir) dir = "tests/golden";
+  char buf[512];
+  snprintf(buf, sizeof(buf), "%s/hf_%016llx.hsaco", dir, (unsigned long long)h);
+  return buf;
+}
+
+// compile the desc's gate kernel; returns the code object (empty on
+// failure). Pure hiprtc — usable without a GPU (the build container
+// pre-compiles and ships the .hsaco files so GPU boxes never pay the
+// ~1.5-3 min compile).
+inline std::vector<char> hf_rtc_compile(const PDesc& d) {
   std::string src = hf_rtc_source(d);
-  if (src.empty()) return nullptr;
+  if (src.empty()) return {};
   hiprtcProgram prog;
   if (hiprtcCreateProgram(&prog, src.c_str(), "hf_gates.hip", 0, nullptr, nullptr) !=
       HIPRTC_SUCCESS)
-    return nullptr;
+    return {};
   const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17"};
   hiprtcResult rc = hiprtcCompileProgram(prog, 3, opts);
   if (rc != HIPRTC_SUCCESS) {
@@ -268,13 +280,47 @@ inline HfRtcKernel* hf_rtc_get(const PDesc& d) {
     fprintf(stderr, "taiga hf_rtc: compile failed, using interpreter:\n%.2000s\n",
             log.c_str());
     hiprtcDestroyProgram(&prog);
-    return nullptr;
+    return {};
   }
   size_t csz = 0;
   hiprtcGetCodeSize(prog, &csz);
   std::vector<char> code(csz);
   hiprtcGetCode(prog, code.data());
   hiprtcDestroyProgram(&prog);
+  return code;
+}
+
+inline HfRtcKernel* hf_rtc_get(const PDesc& d) {
+  static std::mutex mu;
+  static std::map<uint64_t, HfRtcKernel> cache;
+  if (getenv("TG_NO_RTC")) return nullptr;
+  std::lock_guard<std::mutex> lk(mu);
+  uint64_t h = hf_desc_hash(d);
+  auto it = cache.find(h);
+  if (it != cache.end()) return it->second.ready ? &it->second : nullptr;
+  HfRtcKernel& k = cache[h];  // default: not ready (negative-cache failures)
+  std::vector<char> code;
+  // disk cache first (shipped by the build step)
+  {
+    FILE* f = fopen(hf_cache_path(h).c_str(), "rb");
+    if (f) {
+      fseek(f, 0, SEEK_END);
+      long sz = ftell(f);
+      fseek(f, 0, SEEK_SET);
+      code.resize(sz);
+      if (fread(code.data(), 1, sz, f) != (size_t)sz) code.clear();
+      fclose(f);
+    }
+  }
+  if (code.empty()) {
+    code = hf_rtc_compile(d);
+    if (code.empty()) return nullptr;
+    FILE* f = fopen(hf_cache_path(h).c_str(), "wb");
+    if (f) {
+      fwrite(code.data(), 1, code.size(), f);
+      fclose(f);
+    }
+  }
   if (hipModuleLoadData(&k.mod, code.data()) != hipSuccess) return nullptr;
   if (hipModuleGetFunction(&k.fn, k.mod, "hf_gates") != hipSuccess) return nullptr;
   k.ready = true;
