@@ -22,6 +22,7 @@
 #include <hip/hiprtc.h>
 
 #include <cstdio>
+#include <algorithm>
 #include <map>
 #include <mutex>
 #include <array>
@@ -153,12 +154,20 @@ inline std::string hf_rtc_source(const PDesc& d) {
     return tag == XFIXED ? 0 : tag == XADVICE ? d.n_fixed : d.n_fixed + d.n_advice;
   };
 
-  // per-gate function with local query loads + hash-consed temporaries
-  for (size_t gi = 0; gi < d.gates.size(); gi++) {
-    o << "__device__ __noinline__ Fp g" << gi
-      << "(Fp* const* cols, long i, long rs, long mask) {\n";
-    // distinct queries of this gate
+  // constraint clusters (8 per device function): consecutive desc
+  // constraints come from the same halo2 gate, so clustering recovers the
+  // cross-constraint subexpression sharing (complete-add's 12 constraints
+  // share most of their terms) while keeping each function small enough
+  // for sane compile times
+  const size_t GRP = 8;
+  size_t n_groups = (d.gates.size() + GRP - 1) / GRP;
+  for (size_t grp = 0; grp < n_groups; grp++) {
+    size_t g0 = grp * GRP, g1 = std::min(d.gates.size(), g0 + GRP);
+    o << "__device__ __noinline__ Fp g" << grp
+      << "(Fp* const* cols, long i, long rs, long mask, Fp acc, Fp y) {\n";
+    // distinct queries of this cluster
     std::vector<std::array<int, 3>> qs;
+    for (size_t gi = g0; gi < g1; gi++)
     for (const auto& op : d.gates[gi].ops)
       if (op.tag == XFIXED || op.tag == XADVICE || op.tag == XINSTANCE) {
         std::array<int, 3> k{(int)op.tag, (int)op.a, op.b};
@@ -177,10 +186,12 @@ inline std::string hf_rtc_source(const PDesc& d) {
       o << "  const Fp " << qname(q[0], q[1], q[2]) << " = cols["
         << (colbase(q[0]) + q[1]) << "][(i + (long)(" << q[2] << ")*rs) & mask];\n";
     }
-    // postfix -> temporaries with hash-consing (per gate)
+    // postfix -> temporaries with hash-consing (per cluster)
     std::map<std::string, std::string> cse;
     int tmp = 0;
     std::vector<std::string> stk;
+    for (size_t gi = g0; gi < g1; gi++) {
+    stk.clear();
     auto emit = [&](const std::string& expr) {
       auto it = cse.find(expr);
       if (it != cse.end()) return it->second;
@@ -228,7 +239,9 @@ inline std::string hf_rtc_source(const PDesc& d) {
           return std::string();
       }
     }
-    o << "  return " << stk.back() << ";\n}\n";
+    o << "  acc = f_add(f_mul(acc, y), " << stk.back() << ");\n";
+    }  // gi
+    o << "  return acc;\n}\n";
   }
 
   o << "extern \"C\" __global__ void __launch_bounds__(256) hf_gates("
@@ -237,8 +250,8 @@ inline std::string hf_rtc_source(const PDesc& d) {
        "  for (long i = blockIdx.x*(long)blockDim.x + threadIdx.x; i < ext_n;"
        " i += (long)gridDim.x*blockDim.x) {\n"
        "    Fp acc = {{0,0,0,0}};\n";
-  for (size_t gi = 0; gi < d.gates.size(); gi++)
-    o << "    acc = f_add(f_mul(acc, y), g" << gi << "(cols, i, rs, mask));\n";
+  for (size_t grp = 0; grp < n_groups; grp++)
+    o << "    acc = g" << grp << "(cols, i, rs, mask, acc, y);\n";
   o << "    out[i] = acc;\n  }\n}\n";
   return o.str();
 }
