@@ -1093,22 +1093,24 @@ int tg_compliance_prove(tg_ctx* ctx, const uint8_t* info_borsh, size_t len,
   if (d.n_instance_rows != 9 || c->ppk->tgw.n_inputs != 124) return TG_ERR_STATE;
   std::vector<Fp> inputs;
   if (!compliance_inputs(info_borsh, len, inputs)) return TG_ERR_ENCODING;
-  std::vector<uint8_t> advice(32ul * d.n_advice * d.n, 0);
-  if (!c->ppk->tgw.run(inputs, d.n_advice, advice.data())) return TG_ERR_BADARG;
-  uint8_t inst[288];
-  memset(inst, 0, sizeof(inst));
-  {  // anchor row from the blob (instance row 1 = input slot 1)
-    Fp a = fd_from_mont(inputs[1]);
-    memcpy(inst + 32, a.l, 32);
-  }
-  c->ppk->tgw.read_instance(d.n_advice, advice.data(), inst);
+  std::vector<std::vector<Fp>> advice_lag;
+  if (!c->ppk->tgw.run_mont(inputs, d.n_advice, advice_lag)) return TG_ERR_BADARG;
+  std::vector<Fp> inst_rows(9, fd_zero<FpCfg>());
+  inst_rows[1] = inputs[1];  // anchor from the blob
+  c->ppk->tgw.read_instance_mont(advice_lag, inst_rows);
+  std::vector<Fp> inst_lag(d.n, fd_zero<FpCfg>());
+  for (int r = 0; r < 9; r++) inst_lag[r] = inst_rows[r];
   std::vector<uint8_t> proof;
-  int rc = pprove_raw(c, *c->ppk, inst, advice.data(), rng_seed, proof);
+  int rc = pprove_core(c, *c->ppk, inst_lag, advice_lag, rng_seed, proof);
   if (rc != 0) return rc;
   if (proof.size() > cap) return TG_ERR_BADARG;
   memcpy(proof_out, proof.data(), proof.size());
   *out_len = proof.size();
-  if (instance_out) memcpy(instance_out, inst, 288);
+  if (instance_out)
+    for (int r = 0; r < 9; r++) {
+      Fp v = fd_from_mont(inst_rows[r]);
+      memcpy(instance_out + 32 * r, v.l, 32);
+    }
   return TG_OK;
 }
 
@@ -1125,20 +1127,31 @@ int tg_rl_prove(tg_ctx* ctx, const uint8_t* witness_borsh, size_t len,
   PDesc& d = c->ppk->d;
   if (d.n_instance_rows != 22 || c->ppk->tgw.n_inputs != 41) return TG_ERR_STATE;
   std::vector<Fp> inputs;
-  uint8_t inst[704];
-  memset(inst, 0, sizeof(inst));
-  if (!rl_inputs(witness_borsh, len, pad_rseed, inputs, inst + 6 * 32))
+  uint8_t pad[16 * 32];
+  if (!rl_inputs(witness_borsh, len, pad_rseed, inputs, pad))
     return TG_ERR_ENCODING;
-  std::vector<uint8_t> advice(32ul * d.n_advice * d.n, 0);
-  if (!c->ppk->tgw.run(inputs, d.n_advice, advice.data())) return TG_ERR_BADARG;
-  c->ppk->tgw.read_instance(d.n_advice, advice.data(), inst);
+  std::vector<std::vector<Fp>> advice_lag;
+  if (!c->ppk->tgw.run_mont(inputs, d.n_advice, advice_lag)) return TG_ERR_BADARG;
+  std::vector<Fp> inst_rows(22, fd_zero<FpCfg>());
+  for (int r = 0; r < 16; r++) {
+    Fp v;
+    memcpy(v.l, pad + 32 * r, 32);
+    inst_rows[6 + r] = fd_to_mont(v);
+  }
+  c->ppk->tgw.read_instance_mont(advice_lag, inst_rows);
+  std::vector<Fp> inst_lag(d.n, fd_zero<FpCfg>());
+  for (int r = 0; r < 22; r++) inst_lag[r] = inst_rows[r];
   std::vector<uint8_t> proof;
-  int rc = pprove_raw(c, *c->ppk, inst, advice.data(), rng_seed, proof);
+  int rc = pprove_core(c, *c->ppk, inst_lag, advice_lag, rng_seed, proof);
   if (rc != 0) return rc;
   if (proof.size() > cap) return TG_ERR_BADARG;
   memcpy(proof_out, proof.data(), proof.size());
   *out_len = proof.size();
-  if (instance_out) memcpy(instance_out, inst, 704);
+  if (instance_out)
+    for (int r = 0; r < 22; r++) {
+      Fp v = fd_from_mont(inst_rows[r]);
+      memcpy(instance_out + 32 * r, v.l, 32);
+    }
   return TG_OK;
 }
 

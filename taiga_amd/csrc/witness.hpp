@@ -118,6 +118,68 @@ struct TgwProgram {
     return true;
   }
 
+  // Montgomery-direct variant: synthesize straight into per-column Mont
+  // vectors (the prover core's native witness form — skips the canonical
+  // round-trip of run()); exposure values are returned Mont as well.
+  bool run_mont(const std::vector<Fp>& inputs, int n_advice,
+                std::vector<std::vector<Fp>>& advice_lag,
+                std::vector<Fp>* regs_out = nullptr) const {
+    if (inputs.size() != n_inputs) return false;
+    long n = 1L << k;
+    std::vector<Fp> regs(ops.size());
+    for (size_t i = 0; i < ops.size(); i++) {
+      const TgwOp& o = ops[i];
+      switch (o.op) {
+        case W_LOADI: regs[i] = inputs[o.a]; break;
+        case W_CONST: regs[i] = consts[o.a]; break;
+        case W_ADD: regs[i] = fd_add(regs[o.a], regs[o.b]); break;
+        case W_SUB: regs[i] = fd_sub(regs[o.a], regs[o.b]); break;
+        case W_MUL: regs[i] = fd_mul(regs[o.a], regs[o.b]); break;
+        case W_INV0: regs[i] = fd_inv(regs[o.a]); break;
+        case W_NEG: regs[i] = fd_neg(regs[o.a]); break;
+        case W_SQRT0:
+          if (!fd_sqrt(regs[i], regs[o.a])) regs[i] = fd_zero<FpCfg>();
+          break;
+        case W_BIT: {
+          Fp s = fd_from_mont(regs[o.a]);
+          u64 bit = (s.l[o.b >> 6] >> (o.b & 63)) & 1;
+          Fp v{{bit, 0, 0, 0}};
+          regs[i] = fd_to_mont(v);
+          break;
+        }
+        case W_BYTE: {
+          Fp s = fd_from_mont(regs[o.a]);
+          u64 byte = (s.l[o.b >> 3] >> (8 * (o.b & 7))) & 0xFF;
+          Fp v{{byte, 0, 0, 0}};
+          regs[i] = fd_to_mont(v);
+          break;
+        }
+        default:
+          return false;
+      }
+    }
+    advice_lag.assign(n_advice, std::vector<Fp>(n, fd_zero<FpCfg>()));
+    for (size_t s2 = 0; s2 < stores.size(); s2 += 3) {
+      uint32_t col = stores[s2], row = stores[s2 + 1], reg = stores[s2 + 2];
+      if ((int)col >= n_advice || row >= (uint32_t)n || reg >= ops.size())
+        return false;
+      advice_lag[col][row] = regs[reg];
+    }
+    if (regs_out) *regs_out = std::move(regs);
+    return true;
+  }
+
+  // instance rows (Mont) from synthesized Mont columns
+  void read_instance_mont(const std::vector<std::vector<Fp>>& advice_lag,
+                          std::vector<Fp>& inst_rows) const {
+    long n = 1L << k;
+    for (size_t i = 0; i < expose.size(); i += 3) {
+      uint32_t irow = expose[i], col = expose[i + 1], row = expose[i + 2];
+      if (col < advice_lag.size() && row < (uint32_t)n && irow < inst_rows.size())
+        inst_rows[irow] = advice_lag[col][row];
+    }
+  }
+
   // read circuit-computed instance rows from synthesized advice bytes
   void read_instance(int n_advice, const uint8_t* advice,
                      uint8_t* instance_out) const {
