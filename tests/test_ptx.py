@@ -46,7 +46,11 @@ def _build_units():
         tag = b"ptx%d" % i
         rin = mkres(tag + b"in")
         nf = rin.get_nf()
+        # balanced transfer: the output carries the input's kind and
+        # quantity, so sum(delta) = [sum rcv]R and the binding signature
+        # verifies (delta_commitment.rs / transaction.rs:99-114)
         rout = mkres(tag + b"out", nonce=nf)
+        rout.logic, rout.label, rout.quantity = rin.logic, rin.label, rin.quantity
         # commitment-tree path (depth 32) for the compliance proof
         path = [(det(tag + b"n%d" % j, F.P), bool(det(tag + b"l%d" % j, 2)))
                 for j in range(32)]
