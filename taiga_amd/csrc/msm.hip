@@ -219,13 +219,53 @@ __global__ void __launch_bounds__(256) k_scatter(const uint32_t* dig, u64 n, u64
 // serially adding tens of thousands of points is a 100x tail otherwise.
 constexpr uint32_t MSM_BIG_BUCKET = 64;
 
+// ---- bucket length sort (round 2) ----
+// Bucket sizes are Poisson(n/nbuck) (mean ~8 for the prover's c=13 commits),
+// so a 64-lane wave processing 64 random buckets runs at the speed of its
+// LONGEST bucket (E[max of 64 Poisson(8)] ~ 19): ~2.3x of the lane-time is
+// divergence waste. A 66-bin counting sort of bucket ids by length (len
+// 0..64, 65 = big) makes every wave's buckets uniform-length — the same
+// additions in the same per-bucket order, so proofs stay bit-identical.
+constexpr int MSM_LEN_BINS = 66;
+
+__global__ void __launch_bounds__(256) k_len_hist(const uint32_t* start, const uint32_t* end,
+                                                  u64 m, uint32_t* lhist) {
+  for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < m;
+       b += (u64)gridDim.x * blockDim.x) {
+    uint32_t len = end[b] - start[b];
+    atomicAdd(&lhist[len > MSM_BIG_BUCKET ? MSM_LEN_BINS - 1 : len], 1u);
+  }
+}
+
+__global__ void k_len_scan(uint32_t* lhist) {  // exclusive scan, 66 entries
+  if (threadIdx.x == 0) {
+    uint32_t acc = 0;
+    for (int i = 0; i < MSM_LEN_BINS; i++) {
+      uint32_t v = lhist[i];
+      lhist[i] = acc;
+      acc += v;
+    }
+  }
+}
+
+__global__ void __launch_bounds__(256) k_len_scatter(const uint32_t* start, const uint32_t* end,
+                                                     u64 m, uint32_t* loff, uint32_t* order) {
+  for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < m;
+       b += (u64)gridDim.x * blockDim.x) {
+    uint32_t len = end[b] - start[b];
+    uint32_t pos = atomicAdd(&loff[len > MSM_BIG_BUCKET ? MSM_LEN_BINS - 1 : len], 1u);
+    order[pos] = (uint32_t)b;
+  }
+}
+
 template <bool SAFE>
 __global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, const uint32_t* end,
                              const uint32_t* sorted, const VestaAff* pts,
                              VestaJac* buckets, u64 nbuckets_total, uint32_t* big_list,
-                             uint32_t* big_count) {
-  for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < nbuckets_total;
-       b += (u64)gridDim.x * blockDim.x) {
+                             uint32_t* big_count, const uint32_t* order) {
+  for (u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x; t < nbuckets_total;
+       t += (u64)gridDim.x * blockDim.x) {
+    u64 b = order[t];  // length-sorted: waves see uniform bucket sizes
     uint32_t s = start[b], e = end[b];
     if (e - s > MSM_BIG_BUCKET) {
       uint32_t slot = atomicAdd(big_count, 1u);
@@ -402,6 +442,8 @@ struct MsmWork {
   VestaJac* d_partials = nullptr;
   VestaJac* d_wsums = nullptr;  // MSM_NWIN window sums (combined on host)
   uint32_t* d_big = nullptr;    // big-bucket work list + count (phase 2)
+  uint32_t* d_order = nullptr;  // bucket ids counting-sorted by length
+  uint32_t* d_lhist = nullptr;  // MSM_LEN_BINS length histogram / offsets
   u64 cap_n = 0;
   u64 cap_b = 1;
 };
@@ -416,7 +458,7 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n_total, u64 batch = 1) {
   if (p) { hipFree(p); p = nullptr; }
   TGW_FREE(w.d_dig) TGW_FREE(w.d_hist) TGW_FREE(w.d_off) TGW_FREE(w.d_end)
   TGW_FREE(w.d_bsum) TGW_FREE(w.d_sorted) TGW_FREE(w.d_buckets) TGW_FREE(w.d_partials)
-  TGW_FREE(w.d_wsums) TGW_FREE(w.d_big)
+  TGW_FREE(w.d_wsums) TGW_FREE(w.d_big) TGW_FREE(w.d_order) TGW_FREE(w.d_lhist)
 #undef TGW_FREE
   if ((e = hipMalloc(&w.d_dig, n_total * 32 * 4)) != hipSuccess) return e;  /* nwin<=32 (c=8) */
   if ((e = hipMalloc(&w.d_hist, m * 4)) != hipSuccess) return e;
@@ -433,6 +475,8 @@ inline hipError_t msm_work_alloc(MsmWork& w, u64 n_total, u64 batch = 1) {
       hipSuccess)
     return e;
   if ((e = hipMalloc(&w.d_big, (m + 1) * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_order, m * 4)) != hipSuccess) return e;
+  if ((e = hipMalloc(&w.d_lhist, MSM_LEN_BINS * 4)) != hipSuccess) return e;
   w.cap_n = n_total;
   w.cap_b = batch;
   return hipSuccess;
@@ -443,6 +487,17 @@ static inline int msm_grid(u64 work, int block = 256) {
   if (blocks > 2048) blocks = 2048;
   if (blocks < 1) blocks = 1;
   return (int)blocks;
+}
+
+// counting-sort bucket ids by length into w.d_order (see MSM_LEN_BINS note);
+// call between k_scatter (start/end final) and k_bucket_acc.
+inline void msm_len_sort(MsmWork& w, u64 m, hipStream_t stream) {
+  hipMemsetAsync(w.d_lhist, 0, MSM_LEN_BINS * 4, stream);
+  hipLaunchKernelGGL(k_len_hist, dim3(msm_grid(m)), dim3(256), 0, stream, w.d_hist,
+                     w.d_end, m, w.d_lhist);
+  hipLaunchKernelGGL(k_len_scan, dim3(1), dim3(64), 0, stream, w.d_lhist);
+  hipLaunchKernelGGL(k_len_scatter, dim3(msm_grid(m)), dim3(256), 0, stream, w.d_hist,
+                     w.d_end, m, w.d_lhist, w.d_order);
 }
 
 // host-side final combine: acc = sum_w 2^(16w) * wsum[w]  (Horner, ~240
