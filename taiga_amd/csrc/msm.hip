@@ -228,13 +228,22 @@ constexpr uint32_t MSM_BIG_BUCKET = 64;
 // additions in the same per-bucket order, so proofs stay bit-identical.
 constexpr int MSM_LEN_BINS = 66;
 
+// LDS-aggregated: a block counts into LDS and merges 66 counters once —
+// per-item global atomics on 66 addresses measured 1.3 ms/launch (26x the
+// whole sort's useful work) from contention.
 __global__ void __launch_bounds__(256) k_len_hist(const uint32_t* start, const uint32_t* end,
                                                   u64 m, uint32_t* lhist) {
+  __shared__ uint32_t lh[MSM_LEN_BINS];
+  for (int i = threadIdx.x; i < MSM_LEN_BINS; i += blockDim.x) lh[i] = 0;
+  __syncthreads();
   for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < m;
        b += (u64)gridDim.x * blockDim.x) {
     uint32_t len = end[b] - start[b];
-    atomicAdd(&lhist[len > MSM_BIG_BUCKET ? MSM_LEN_BINS - 1 : len], 1u);
+    atomicAdd(&lh[len > MSM_BIG_BUCKET ? MSM_LEN_BINS - 1 : len], 1u);
   }
+  __syncthreads();
+  for (int i = threadIdx.x; i < MSM_LEN_BINS; i += blockDim.x)
+    if (lh[i]) atomicAdd(&lhist[i], lh[i]);
 }
 
 __global__ void k_len_scan(uint32_t* lhist) {  // exclusive scan, 66 entries
@@ -248,13 +257,36 @@ __global__ void k_len_scan(uint32_t* lhist) {  // exclusive scan, 66 entries
   }
 }
 
+// Tiled two-phase placement: a block ranks a 2048-bucket tile in LDS, then
+// reserves per-bin ranges with ONE global atomic per live bin. Order within
+// a bin is racy across tiles — irrelevant: every bucket is processed once
+// and results are stored by bucket id, so proof bytes are unaffected.
 __global__ void __launch_bounds__(256) k_len_scatter(const uint32_t* start, const uint32_t* end,
                                                      u64 m, uint32_t* loff, uint32_t* order) {
-  for (u64 b = blockIdx.x * (u64)blockDim.x + threadIdx.x; b < m;
-       b += (u64)gridDim.x * blockDim.x) {
-    uint32_t len = end[b] - start[b];
-    uint32_t pos = atomicAdd(&loff[len > MSM_BIG_BUCKET ? MSM_LEN_BINS - 1 : len], 1u);
-    order[pos] = (uint32_t)b;
+  __shared__ uint32_t cnt[MSM_LEN_BINS];
+  __shared__ uint32_t base[MSM_LEN_BINS];
+  const u64 tile_sz = (u64)blockDim.x * 8;
+  for (u64 t0 = (u64)blockIdx.x * tile_sz; t0 < m; t0 += (u64)gridDim.x * tile_sz) {
+    for (int i = threadIdx.x; i < MSM_LEN_BINS; i += blockDim.x) cnt[i] = 0;
+    __syncthreads();
+    uint32_t bin[8], rank[8];
+    for (int j = 0; j < 8; j++) {
+      u64 b = t0 + (u64)j * blockDim.x + threadIdx.x;
+      if (b < m) {
+        uint32_t len = end[b] - start[b];
+        bin[j] = len > MSM_BIG_BUCKET ? MSM_LEN_BINS - 1 : len;
+        rank[j] = atomicAdd(&cnt[bin[j]], 1u);
+      }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < MSM_LEN_BINS; i += blockDim.x)
+      base[i] = cnt[i] ? atomicAdd(&loff[i], cnt[i]) : 0u;
+    __syncthreads();
+    for (int j = 0; j < 8; j++) {
+      u64 b = t0 + (u64)j * blockDim.x + threadIdx.x;
+      if (b < m) order[base[bin[j]] + rank[j]] = (uint32_t)b;
+    }
+    __syncthreads();
   }
 }
 
@@ -265,7 +297,8 @@ __global__ void __launch_bounds__(256, 1) k_bucket_acc(const uint32_t* start, co
                              uint32_t* big_count, const uint32_t* order) {
   for (u64 t = blockIdx.x * (u64)blockDim.x + threadIdx.x; t < nbuckets_total;
        t += (u64)gridDim.x * blockDim.x) {
-    u64 b = order[t];  // length-sorted: waves see uniform bucket sizes
+    // length-sorted order (null = identity): waves see uniform bucket sizes
+    u64 b = order ? order[t] : t;
     uint32_t s = start[b], e = end[b];
     if (e - s > MSM_BIG_BUCKET) {
       uint32_t slot = atomicAdd(big_count, 1u);
@@ -490,14 +523,28 @@ static inline int msm_grid(u64 work, int block = 256) {
 }
 
 // counting-sort bucket ids by length into w.d_order (see MSM_LEN_BINS note);
-// call between k_scatter (start/end final) and k_bucket_acc.
-inline void msm_len_sort(MsmWork& w, u64 m, hipStream_t stream) {
+// call between k_scatter (start/end final) and k_bucket_acc. Returns the
+// order array to hand to k_bucket_acc, or null when the sort is skipped:
+// only the SMALL-window configs (the prover's n=2^15 commits, c<=13, mean
+// bucket ~8: wave time = max-of-64 Poisson(8) ~ 2.3x mean) benefit; at
+// c=16/n=2^20 (mean 32, tight lengths) the indirection scatters the
+// otherwise-coalesced sorted[]-run reads and bucket writes and measured
+// 25% SLOWER — so large windows keep the identity order.
+// TG_MSM_LEN_SORT=0/1 forces it off/on for A/B probes.
+inline const uint32_t* msm_len_sort(MsmWork& w, const MsmCfg& cfg, u64 m,
+                                    hipStream_t stream) {
+  static int force = [] {
+    const char* e = getenv("TG_MSM_LEN_SORT");
+    return e ? (atoi(e) ? 1 : 0) : -1;
+  }();
+  if (force == 0 || (force == -1 && cfg.c > 13)) return nullptr;
   hipMemsetAsync(w.d_lhist, 0, MSM_LEN_BINS * 4, stream);
   hipLaunchKernelGGL(k_len_hist, dim3(msm_grid(m)), dim3(256), 0, stream, w.d_hist,
                      w.d_end, m, w.d_lhist);
   hipLaunchKernelGGL(k_len_scan, dim3(1), dim3(64), 0, stream, w.d_lhist);
-  hipLaunchKernelGGL(k_len_scatter, dim3(msm_grid(m)), dim3(256), 0, stream, w.d_hist,
-                     w.d_end, m, w.d_lhist, w.d_order);
+  hipLaunchKernelGGL(k_len_scatter, dim3(msm_grid(m, 256 * 8)), dim3(256), 0, stream,
+                     w.d_hist, w.d_end, m, w.d_lhist, w.d_order);
+  return w.d_order;
 }
 
 // host-side final combine: acc = sum_w 2^(16w) * wsum[w]  (Horner, ~240

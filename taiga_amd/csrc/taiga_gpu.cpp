@@ -494,20 +494,18 @@ static int msm_common(Ctx* c, size_t n, int base_set, uint8_t out_xy[64]) {
     {
       ProfScope p(c, P_MSM_ACC);
       hipMemsetAsync(c->msm.d_big + m, 0, 4, c->stream);
-      msm_len_sort(c->msm, m, c->stream);
+      const uint32_t* ord = msm_len_sort(c->msm, cfg, m, c->stream);
       if (safe) {
         hipLaunchKernelGGL(k_bucket_acc<true>, dim3(msm_grid(m)), dim3(256), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
-                           c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m,
-                           c->msm.d_order);
+                           c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m, ord);
         hipLaunchKernelGGL(k_bucket_acc_big<true>, dim3(1024), dim3(MSM_BIG_LANES), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
                            c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);
       } else {
         hipLaunchKernelGGL(k_bucket_acc<false>, dim3(msm_grid(m)), dim3(256), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
-                           c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m,
-                           c->msm.d_order);
+                           c->msm.d_buckets, m, c->msm.d_big, c->msm.d_big + m, ord);
         hipLaunchKernelGGL(k_bucket_acc_big<false>, dim3(1024), dim3(MSM_BIG_LANES), 0, c->stream,
                            c->msm.d_hist, c->msm.d_end, c->msm.d_sorted, bases,
                            c->msm.d_buckets, c->msm.d_big, c->msm.d_big + m);

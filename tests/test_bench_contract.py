@@ -14,8 +14,18 @@ REQUIRED = ["metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step
             "config", "roofline", "cpu_baseline"]
 
 
-@pytest.mark.parametrize("name", ["r01_final_proof", "r01_final_msm",
-                                  "r01_final_ntt", "r01_final_verify"])
+def _final_lines():
+    import glob
+
+    names = {os.path.splitext(os.path.basename(p))[0]
+             for p in glob.glob(os.path.join(PROFILES, "r0*_final_*.json"))}
+    # the round-1 four must always exist; later rounds' lines join the sweep
+    names |= {"r01_final_proof", "r01_final_msm", "r01_final_ntt",
+              "r01_final_verify"}
+    return sorted(names)
+
+
+@pytest.mark.parametrize("name", _final_lines())
 def test_bench_line_schema(name):
     path = os.path.join(PROFILES, f"{name}.json")
     d = json.loads(open(path).read().strip().splitlines()[-1])
