@@ -524,20 +524,20 @@ static inline int msm_grid(u64 work, int block = 256) {
 
 // counting-sort bucket ids by length into w.d_order (see MSM_LEN_BINS note);
 // call between k_scatter (start/end final) and k_bucket_acc. Returns the
-// order array to hand to k_bucket_acc, or null when the sort is skipped:
-// only the SMALL-window configs (the prover's n=2^15 commits, c<=13, mean
-// bucket ~8: wave time = max-of-64 Poisson(8) ~ 2.3x mean) benefit; at
-// c=16/n=2^20 (mean 32, tight lengths) the indirection scatters the
-// otherwise-coalesced sorted[]-run reads and bucket writes and measured
-// 25% SLOWER — so large windows keep the identity order.
-// TG_MSM_LEN_SORT=0/1 forces it off/on for A/B probes.
+// order array to hand to k_bucket_acc (null = sort disabled via
+// TG_MSM_LEN_SORT=0, A/B only). Measured on MI355X: 1.44x on k_bucket_acc
+// at the prover's c=13 (Poisson(8) buckets, wave runs at max-of-64 ~ 19
+// without it) and 1.15x at c=16/n=2^20 — but ONLY with the LDS-aggregated
+// histogram/scatter below; the first cut's per-item global atomics on 66
+// bins serialized the whole chip across all proving streams.
 inline const uint32_t* msm_len_sort(MsmWork& w, const MsmCfg& cfg, u64 m,
                                     hipStream_t stream) {
   static int force = [] {
     const char* e = getenv("TG_MSM_LEN_SORT");
     return e ? (atoi(e) ? 1 : 0) : -1;
   }();
-  if (force == 0 || (force == -1 && cfg.c > 13)) return nullptr;
+  if (force == 0) return nullptr;
+  (void)cfg;
   hipMemsetAsync(w.d_lhist, 0, MSM_LEN_BINS * 4, stream);
   hipLaunchKernelGGL(k_len_hist, dim3(msm_grid(m)), dim3(256), 0, stream, w.d_hist,
                      w.d_end, m, w.d_lhist);
