@@ -500,7 +500,7 @@ def main():
                 "unit": unit,
                 "cores": cores,
                 "kind": "port",
-                "sample": "one CS1 proof verification via the oracle (single check)",
+                "sample": "one exact-compliance proof verification via the oracle (single check)",
             }
         elif args.workload == "msm":
             nb = 1 << 17  # bounded sample (~10-30 s of CPU work)
