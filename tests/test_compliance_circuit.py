@@ -103,6 +103,11 @@ def test_oracle_prove_verify(oracle, name, size):
     out = ctypes.create_string_buffer(1 << 16)
     plen = oracle.orc_prove_raw(inst, adv, RNG, out, ctypes.c_long(1 << 16))
     assert plen == size
+    # frozen absolute-bytes pin (tools/gen_proof_fixture.py): catches
+    # shared-design drift that would move oracle AND GPU together (the
+    # GPU tier's bit-identity then extends the pin to the product)
+    pin = open(os.path.join(GOLDEN, f"{name}_proof_pin.bin"), "rb").read()
+    assert out.raw[:plen] == pin, f"{name}: proof bytes drifted from the pin"
     assert oracle.orc_verify_raw(inst, out, ctypes.c_long(plen)) == 0
     bad = bytearray(out.raw[:plen])
     bad[200] ^= 1
