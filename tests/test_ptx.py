@@ -22,7 +22,7 @@ sys.path.insert(0, os.path.join(REPO, "tools"))
 RNG = bytes([17]) + bytes(31)
 
 
-def _build_units():
+def _build_units(n_compliance=2):
     from circuit import fields as F
     from circuit import hostcrypto as hc
 
@@ -42,7 +42,7 @@ def _build_units():
     rl_wits_in, rl_wits_out = [], []
     leaves = []
     resources = []
-    for i in range(2):
+    for i in range(n_compliance):
         tag = b"ptx%d" % i
         rin = mkres(tag + b"in")
         nf = rin.get_nf()
@@ -68,7 +68,8 @@ def _build_units():
         leaves += [nf, rout.commitment()]
         resources.append((rin, rout))
 
-    # resource tree: depth 4, leaves padded with zeros
+    # resource tree: depth 4, leaves padded with zeros (ragged for
+    # n_compliance != 2: 2*n leaves used of 16)
     layer = leaves + [0] * (16 - len(leaves))
     layers = [layer]
     while len(layer) > 1:
@@ -127,8 +128,62 @@ def _orc_ptx(lib, comp, rin, rout, out):
         cdesc, ctypes.c_long(len(cdesc)), rdesc, ctypes.c_long(len(rdesc)),
         srs, ctypes.c_long(len(srs)), ctgw, ctypes.c_long(len(ctgw)),
         rtgw, ctypes.c_long(len(rtgw)),
-        2, b"".join(comp), 2, 2, b"".join(rin) + b"".join(rout),
+        len(comp), b"".join(comp), len(rin), len(rout),
+        b"".join(rin) + b"".join(rout),
         RNG, out, ctypes.c_long(len(out)))
+
+
+@pytest.mark.parametrize("n", [1])
+def test_oracle_ptx_ragged_sizes(n):
+    """ragged bundle (1 compliance unit; 2 of 16 resource-tree leaves in
+    use): the oracle builds and structure-checks it — the SAME padding
+    path the reference's resource tree zero-fills (resource_tree.rs
+    ResourceMerkleTreeLeaves). n=3 runs in the GPU tier where the proofs
+    are cheap (the CPU suite stays within its minutes budget)."""
+    import struct
+
+    lib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+    lib.orc_ptx_build.restype = ctypes.c_long
+    comp, rin, rout = _build_units(n)
+    out = ctypes.create_string_buffer(1 << 19)
+    m = _orc_ptx(lib, comp, rin, rout, out)
+    assert m > 0, f"orc_ptx_build(n={n}) failed: {m}"
+    (nc,) = struct.unpack_from("<I", out.raw, 0)
+    assert nc == n
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n", [1, 3])
+def test_gpu_ptx_ragged_parity(n, params15):
+    """GPU == oracle bit-for-bit on ragged bundles too (1 and 3
+    compliance units), and the bundles verify."""
+    import taiga_amd
+
+    comp, rin, rout = _build_units(n)
+    g = taiga_amd.TaigaGpu(0)
+    try:
+        g.load_srs(params15)
+        slot_c = g.keygen(open(os.path.join(GOLDEN, "compliance.desc"), "rb").read())
+        g.witness_program_load(open(os.path.join(GOLDEN, "compliance.tgw"), "rb").read())
+        slot_r = g.keygen(open(os.path.join(GOLDEN, "trivial_rl.desc"), "rb").read())
+        g.witness_program_load(open(os.path.join(GOLDEN, "trivial_rl.tgw"), "rb").read())
+        lib = taiga_amd.api.load_library()
+        out = ctypes.create_string_buffer(1 << 19)
+        out_len = ctypes.c_size_t()
+        rc = lib.tg_ptx_build(g._h, slot_c, slot_r, n, b"".join(comp),
+                              n, n, b"".join(rin) + b"".join(rout), RNG, out,
+                              len(out), ctypes.byref(out_len))
+        assert rc == 0, f"tg_ptx_build(n={n}) rc={rc}"
+        ptx_gpu = out.raw[:out_len.value]
+        olib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+        olib.orc_ptx_build.restype = ctypes.c_long
+        oout = ctypes.create_string_buffer(1 << 19)
+        m = _orc_ptx(olib, comp, rin, rout, oout)
+        assert m == len(ptx_gpu)
+        assert oout.raw[:m] == ptx_gpu, f"n={n}: GPU bundle != oracle bundle"
+        assert lib.tg_ptx_verify(g._h, slot_c, slot_r, ptx_gpu, len(ptx_gpu)) == 0
+    finally:
+        g.close()
 
 
 @pytest.mark.gpu
