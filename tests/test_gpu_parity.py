@@ -124,6 +124,39 @@ def test_msm_edge_cases(gpu):
     assert gpu.msm(b"\x00" * (32 * n), base_set=0) == b"\x00" * 64
 
 
+def test_msm_linearity_2e20(gpu):
+    """full BASELINE configs[1] size (2^20): size-independent property —
+    MSM(a) + MSM(b) = MSM(a+b mod p) over the same device-generated base
+    set (linearity; elementwise oracle parity is pinned at
+    oracle-checkable sizes above). Host-side point add via pypasta."""
+    n = 1 << 20
+    gpu.gen_bases(n, seed=42)  # synthetic distinct bases on device
+    rng = random.Random(99)
+    # vectorized scalar generation (python-int loop at 2^20 is too slow)
+    import numpy as np
+
+    raw = np.frombuffer(rng.getrandbits(2 * n * 256).to_bytes(2 * n * 32, "little"),
+                        dtype=np.uint8).copy()
+    raw[31::32] &= 0x3F  # < 2^254 < p: canonical
+    a, b = raw[: n * 32].tobytes(), raw[n * 32:].tobytes()
+    arr = np.frombuffer(raw, dtype="<u8").reshape(2 * n, 4).astype(object)
+    vals = arr[:, 0] + (arr[:, 1] << 64) + (arr[:, 2] << 128) + (arr[:, 3] << 192)
+    s = (vals[:n] + vals[n:]) % pp.P
+    ab = b"".join(int(v).to_bytes(32, "little") for v in s)
+    ra, rb, rab = gpu.msm(a, base_set=0), gpu.msm(b, base_set=0), gpu.msm(ab, base_set=0)
+
+    def pt(r):
+        x = int.from_bytes(r[:32], "little")
+        y = int.from_bytes(r[32:], "little")
+        return None if x == 0 and y == 0 else pp.Point(x, y, pp.Q)
+
+    pa, pb, pab = pt(ra), pt(rb), pt(rab)
+    assert pa is not None and pb is not None and pab is not None
+    assert pa.is_on_curve() and pab.is_on_curve()
+    got = pa + pb
+    assert (got.x, got.y) == (pab.x, pab.y), "MSM linearity broken at 2^20"
+
+
 def test_msm_rejects_bad_point(gpu):
     import taiga_amd
 
