@@ -279,3 +279,67 @@ def test_gpu_full_transaction_verify(units, params15):
     bad2[-1] ^= 1
     assert lib.tg_tx_verify_full(g._h, slot_c, slot_r, bytes(bad2), len(bad2)) != 0
     g.close()
+
+
+@pytest.mark.gpu
+def test_gpu_two_ptx_transaction_verify(params15):
+    """A Transaction aggregating TWO ShieldedPartialTransactions: the
+    binding signing key is the sum of the per-ptx binding_sig_r scalars
+    and the digest streams span both bundles (transaction.rs:99-158 —
+    Transaction::sign over all ptxs)."""
+    import struct
+
+    import taiga_amd
+    from taiga_amd import wire
+
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import pypasta as pp
+
+    g = taiga_amd.TaigaGpu(0)
+    try:
+        g.load_srs(params15)
+        slot_c = g.keygen(open(os.path.join(GOLDEN, "compliance.desc"), "rb").read())
+        g.witness_program_load(open(os.path.join(GOLDEN, "compliance.tgw"), "rb").read())
+        slot_r = g.keygen(open(os.path.join(GOLDEN, "trivial_rl.desc"), "rb").read())
+        g.witness_program_load(open(os.path.join(GOLDEN, "trivial_rl.tgw"), "rb").read())
+        lib = taiga_amd.api.load_library()
+
+        ptxs, rs = [], []
+        for n, rng0 in ((1, 31), (2, 32)):
+            comp, rin, rout = _build_units(n)
+            out = ctypes.create_string_buffer(1 << 19)
+            out_len = ctypes.c_size_t()
+            rc = lib.tg_ptx_build(g._h, slot_c, slot_r, n, b"".join(comp),
+                                  n, n, b"".join(rin) + b"".join(rout),
+                                  bytes([rng0]) + bytes(31), out, len(out),
+                                  ctypes.byref(out_len))
+            assert rc == 0
+            ptx = out.raw[:out_len.value]
+            assert ptx[-37] == 1
+            rs.append(int.from_bytes(ptx[-36:-4], "little"))
+            ptxs.append(ptx)
+
+        nfs, cms, deltas, anchors = [], [], [], []
+        for ptx in ptxs:
+            (n_cvi,) = struct.unpack_from("<I", ptx, 0)
+            off = 4
+            for _ in range(n_cvi):
+                (plen,) = struct.unpack_from("<I", ptx, off)
+                off += 4 + plen
+                inst = ptx[off:off + 192]
+                off += 192
+                anchors.append(inst[0:32])
+                nfs.append(inst[32:64])
+                cms.append(inst[64:96])
+                deltas.append(inst[96:128])
+        digest = taiga_amd.tx_digest(nfs, cms, deltas, anchors)
+        r_total = ((rs[0] + rs[1]) % pp.Q).to_bytes(32, "little")
+        sig = taiga_amd.binding_sign(r_total, digest, bytes([6]) * 32)
+        tx = wire.transaction(ptxs, sig)
+        rc = lib.tg_tx_verify_full(g._h, slot_c, slot_r, tx, len(tx))
+        assert rc == 0, f"two-ptx tg_tx_verify_full rc={rc}"
+        bad = bytearray(tx)
+        bad[-1] ^= 1
+        assert lib.tg_tx_verify_full(g._h, slot_c, slot_r, bytes(bad), len(bad)) != 0
+    finally:
+        g.close()
