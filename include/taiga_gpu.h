@@ -75,6 +75,9 @@ int tg_srs_k(const tg_ctx* ctx); /* k, or TG_ERR_NOSRS */
 int tg_bases_upload(tg_ctx* ctx, const uint8_t* points_xy, size_t n);
 /* synthetic distinct bases generated on-device ([seed+i+1]G) for benches */
 int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed);
+/* read back the current base set as n x 64B canonical affine (test/debug:
+ * lets parity tests diff device-generated bases against the oracle's) */
+int tg_bases_download(tg_ctx* ctx, size_t n, uint8_t* out_xy);
 int tg_msm_pallas(tg_ctx* ctx, const uint8_t* scalars, size_t n, int base_set,
                   uint8_t out_xy[64]);
 

@@ -53,6 +53,8 @@ def load_library():
         lib.tg_bases_upload.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
         lib.tg_scalars_upload.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t]
         lib.tg_gen_bases.argtypes = [ctypes.c_void_p, ctypes.c_size_t, ctypes.c_uint64]
+        lib.tg_bases_download.argtypes = [ctypes.c_void_p, ctypes.c_size_t,
+                                          ctypes.c_char_p]
         lib.tg_msm_pallas.argtypes = [
             ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_int, ctypes.c_char_p,
         ]
@@ -231,6 +233,11 @@ class TaigaGpu:
 
     def gen_bases(self, n: int, seed: int = 0):
         self._ck(self._lib.tg_gen_bases(self._h, n, seed))
+
+    def bases_download(self, n: int) -> bytes:
+        out = ctypes.create_string_buffer(64 * n)
+        self._ck(self._lib.tg_bases_download(self._h, n, out))
+        return out.raw
 
     def msm(self, scalars: bytes, base_set=0) -> bytes:
         out = ctypes.create_string_buffer(64)

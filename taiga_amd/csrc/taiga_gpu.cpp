@@ -437,6 +437,22 @@ int tg_gen_bases(tg_ctx* ctx, size_t n, uint64_t seed) {
   return TG_OK;
 }
 
+int tg_bases_download(tg_ctx* ctx, size_t n, uint8_t* out_xy) {
+  Ctx* c = (Ctx*)ctx;
+  if (!c || tg_enter(c)) return TG_ERR_BADARG;
+  if (!n || !c->d_bases || c->n_bases < n) return TG_ERR_BADARG;
+  hipError_t e;
+  Fq* tmp = nullptr;
+  if ((e = hipMalloc(&tmp, n * sizeof(VestaAff))) != hipSuccess)
+    return set_err(c, "bases dl alloc", e);
+  hipLaunchKernelGGL(k_from_mont<FqCfg>, dim3(ntt_grid(2 * n)), dim3(256), 0, c->stream,
+                     tmp, (const Fq*)c->d_bases, (u64)(2 * n));
+  hipMemcpyAsync(out_xy, tmp, n * 64, hipMemcpyDeviceToHost, c->stream);
+  e = hipStreamSynchronize(c->stream);
+  hipFree(tmp);
+  return e == hipSuccess ? TG_OK : set_err(c, "bases dl", e);
+}
+
 int tg_scalars_upload(tg_ctx* ctx, const uint8_t* scalars, size_t n) {
   Ctx* c = (Ctx*)ctx;
   if (!c || tg_enter(c)) return TG_ERR_BADARG;
