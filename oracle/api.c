@@ -112,9 +112,17 @@ int orc_pt_on_curve(int fid, const uint8_t* a) {
     return fd_eq(lhs, rhs);
 }
 
-/* generate the bench's synthetic distinct bases: out[i] = [seed+i+1]G on
- * Vesta, 64-byte canonical affine — same definition as the product's
- * tg_gen_bases (used by bench.py's cpu_baseline leg and by tests). */
+/* generate the bench's synthetic distinct bases: out[i] = [k_i]G on
+ * Vesta with k_i a splitmix64-derived 256-bit scalar, 64-byte canonical
+ * affine — byte-identical to the product's tg_gen_bases (msm.hip
+ * k_gen_bases; see the note there on why the multiples must be RANDOM,
+ * not small/sequential). Used by bench.py's cpu_baseline leg and tests. */
+static uint64_t orc_sm64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
 void orc_gen_bases(long n, uint64_t seed, uint8_t* out) {
     const fd_ctx* f = &FD_Q;
 #ifdef _OPENMP
@@ -127,7 +135,10 @@ void orc_gen_bases(long n, uint64_t seed, uint8_t* out) {
         fd_neg(G.x, one, f);
         fd_add(G.y, one, one, f);
         G.inf = 0;
-        uint64_t k[4] = {seed + (uint64_t)i + 1, 0, 0, 0};
+        uint64_t k[4];
+        for (int l = 0; l < 4; l++)
+            k[l] = orc_sm64(seed * 0xD1B54A32D192ED03ULL + (uint64_t)i * 4 +
+                            (uint64_t)l);
         pt_jac j, g;
         pt_from_aff(&g, &G, f);
         pt_mul(&j, &g, k, f);

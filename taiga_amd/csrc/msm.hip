@@ -439,9 +439,21 @@ __global__ void __launch_bounds__(256) k_wsum(const VestaJac* partials, VestaJac
   if (t == 0) wsums[w] = lds[0];
 }
 
-// synthetic bench bases: out[i] = [seed + i + 1] * G (distinct points; the
-// payload values don't affect Pippenger's work profile — digits come from
-// the scalars). Per-thread short double-and-add + one field inversion.
+// synthetic bench bases: out[i] = [k_i] G with k_i a splitmix64-derived
+// 256-bit scalar — RANDOM multiples, like the SRS points the production
+// MSMs gather. Small sequential multiples ([seed+i+1]G, the first cut) are
+// NOT usable with the fast accumulation path: bucket partial sums are
+// [sum of +-k]G with |sum| < 2^27, which collides with the next point
+// [+-k_j]G at small-INTEGER rates — the doubling/annihilation cases the
+// fast kernel omits (caught by the 2^20 MSM linearity property test,
+// round 2; the prover was never exposed: SRS bases are random points).
+// Deterministic and byte-identical to the oracle's orc_gen_bases.
+__device__ __forceinline__ u64 tg_sm64(u64 x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
 __global__ void __launch_bounds__(256) k_gen_bases(VestaAff* out, u64 n, u64 seed) {
   // generator (-1, 2) in Mont form
   for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
@@ -451,13 +463,14 @@ __global__ void __launch_bounds__(256) k_gen_bases(VestaAff* out, u64 n, u64 see
     G.x = fd_neg(one);
     Fq two = fd_add(one, one);
     G.y = two;
-    u64 k = seed + i + 1;
+    u64 k[4];
+    for (int j = 0; j < 4; j++)
+      k[j] = tg_sm64(seed * 0xD1B54A32D192ED03ull + i * 4 + (u64)j);
     VestaJac acc = jac_identity<FqCfg>();
     VestaJac base = jac_from_aff(G);
-    while (k) {
-      if (k & 1) acc = jac_add(acc, base);
+    for (int b = 0; b < 256; b++) {
+      if ((k[b >> 6] >> (b & 63)) & 1) acc = jac_add(acc, base);
       base = jac_dbl(base);
-      k >>= 1;
     }
     out[i] = jac_to_aff(acc);
   }

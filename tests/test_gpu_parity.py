@@ -124,6 +124,17 @@ def test_msm_edge_cases(gpu):
     assert gpu.msm(b"\x00" * (32 * n), base_set=0) == b"\x00" * 64
 
 
+def test_gen_bases_parity(gpu):
+    """device-generated synthetic bases == the oracle's, byte for byte
+    (splitmix64-derived RANDOM 256-bit multiples — small sequential
+    multiples made bucket partial sums collide with upcoming points at
+    integer rates, violating the fast accumulation kernel's
+    no-exceptional-case assumption; found by test_msm_linearity_2e20)."""
+    n = 1 << 16
+    gpu.gen_bases(n, seed=42)
+    assert gpu.bases_download(n) == oc.gen_bases(n, 42)
+
+
 def test_msm_linearity_2e20(gpu):
     """full BASELINE configs[1] size (2^20): size-independent property —
     MSM(a) + MSM(b) = MSM(a+b mod p) over the same device-generated base
