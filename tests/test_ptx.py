@@ -133,25 +133,6 @@ def _orc_ptx(lib, comp, rin, rout, out):
         RNG, out, ctypes.c_long(len(out)))
 
 
-@pytest.mark.parametrize("n", [1])
-def test_oracle_ptx_ragged_sizes(n):
-    """ragged bundle (1 compliance unit; 2 of 16 resource-tree leaves in
-    use): the oracle builds and structure-checks it — the SAME padding
-    path the reference's resource tree zero-fills (resource_tree.rs
-    ResourceMerkleTreeLeaves). n=3 runs in the GPU tier where the proofs
-    are cheap (the CPU suite stays within its minutes budget)."""
-    import struct
-
-    lib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
-    lib.orc_ptx_build.restype = ctypes.c_long
-    comp, rin, rout = _build_units(n)
-    out = ctypes.create_string_buffer(1 << 19)
-    m = _orc_ptx(lib, comp, rin, rout, out)
-    assert m > 0, f"orc_ptx_build(n={n}) failed: {m}"
-    (nc,) = struct.unpack_from("<I", out.raw, 0)
-    assert nc == n
-
-
 @pytest.mark.gpu
 @pytest.mark.parametrize("n", [1, 3])
 def test_gpu_ptx_ragged_parity(n, params15):
