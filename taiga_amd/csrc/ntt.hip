@@ -286,6 +286,26 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
       [[maybe_unused]] auto sc = prof(1);
       u64 ntiles = n >> f;
       unsigned grid = ntiles > 2048 ? 2048 : (unsigned)ntiles;
+      // wider residue blocks for the F=7 passes (512-B / 1-KiB contiguous
+      // gathers; gfx950 allows the 66/132 KiB LDS workgroups). Default
+      // RB=16, measured faster than 8 at k=19/22; TG_NTT_RB={8,16,32}
+      // overrides for A/B probes. span >= 512 in every F=7 pass, so the
+      // span%RB constraint is always met here.
+      static int rb_wide = [] {
+        const char* e = getenv("TG_NTT_RB");
+        int v = e ? atoi(e) : 16;
+        return (v == 8 || v == 16 || v == 32) ? v : 16;
+      }();
+      if (f == 7 && rb_wide == 32 && span >= 32) {
+        hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 7, 32>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1);
+        s += f;
+        continue;
+      }
+      if (f == 7 && rb_wide == 16 && span >= 16) {
+        hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 7, 16>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1);
+        s += f;
+        continue;
+      }
       switch (f) {
         case 7: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 7, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
         case 6: hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 6, RB>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1); break;
