@@ -287,14 +287,16 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
       u64 ntiles = n >> f;
       unsigned grid = ntiles > 2048 ? 2048 : (unsigned)ntiles;
       // wider residue blocks for the F=7 passes (512-B / 1-KiB contiguous
-      // gathers; gfx950 allows the 66/132 KiB LDS workgroups). Default
-      // RB=16, measured faster than 8 at k=19/22; TG_NTT_RB={8,16,32}
-      // overrides for A/B probes. span >= 512 in every F=7 pass, so the
-      // span%RB constraint is always met here.
+      // gathers; gfx950 allows the 66/132 KiB LDS workgroups) — MEASURED
+      // SLOWER at k=22 (RB=8: 4.20 G elem/s, 16: 3.95, 32: 3.35): the
+      // 66/132 KiB workgroups drop residency to 2/1 per CU and the lost
+      // latency hiding outweighs the wider gathers, so 256-B chunks were
+      // not the bottleneck. Kept behind TG_NTT_RB={8,16,32} as the A/B
+      // evidence; default stays 8.
       static int rb_wide = [] {
         const char* e = getenv("TG_NTT_RB");
-        int v = e ? atoi(e) : 16;
-        return (v == 8 || v == 16 || v == 32) ? v : 16;
+        int v = e ? atoi(e) : 8;
+        return (v == 8 || v == 16 || v == 32) ? v : 8;
       }();
       if (f == 7 && rb_wide == 32 && span >= 32) {
         hipLaunchKernelGGL((k_ntt_fused2<FpCfg, 7, 32>), dim3(grid), dim3(256), 0, stream, d_tmp, tw, k, s - 1);
