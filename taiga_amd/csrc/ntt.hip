@@ -126,6 +126,46 @@ __global__ void __launch_bounds__(256) k_ntt_fused(Fd<C>* a, const Fd<C>* tw, in
   }
 }
 
+// first-pass variant with the bit-reversal fused into the load: reads the
+// NATURAL-order input directly (a strided gather — source index =
+// bitrev_FUSE(m) << (k-FUSE) | bitrev_{k-FUSE}(tile)) and writes the
+// pass-1 result contiguously, replacing the separate k_bitrev_load sweep
+// (one full n-read + n-write saved per direction).
+template <class C, int FUSE>
+__global__ void __launch_bounds__(256) k_ntt_fused_br(Fd<C>* out, const Fd<C>* in,
+                                                      const Fd<C>* tw, int k) {
+  constexpr int TILE = 1 << FUSE;
+  __shared__ Fd<C> lds[1 << FUSE];
+  u64 n = 1ULL << k;
+  u64 ntiles = n >> FUSE;
+  int kshift = k - FUSE;
+  for (u64 tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    u64 base = tile << FUSE;
+    u64 src_off = bitrev(tile, kshift);
+    for (int m = threadIdx.x; m < TILE; m += blockDim.x)
+      lds[m] = in[(bitrev((u64)m, FUSE) << kshift) | src_off];
+    __syncthreads();
+    for (int ls = 1; ls <= FUSE; ls++) {
+      int s = ls;  // s0 = 0, r = 0
+      u64 half = 1ULL << (ls - 1);
+      int tshift = k - s;
+      for (int j = threadIdx.x; j < (TILE >> 1); j += blockDim.x) {
+        u64 grp = (u64)j >> (ls - 1);
+        u64 kk = (u64)j & (half - 1);
+        u64 i0 = (grp << ls) | kk;
+        u64 i1 = i0 + half;
+        Fd<C> t = fd_mul(lds[i1], tw[kk << tshift]);
+        Fd<C> lo = lds[i0];
+        lds[i1] = fd_sub(lo, t);
+        lds[i0] = fd_add(lo, t);
+      }
+      __syncthreads();
+    }
+    for (int m = threadIdx.x; m < TILE; m += blockDim.x) out[base + (u64)m] = lds[m];
+    __syncthreads();
+  }
+}
+
 // 2-D fused LDS stages for strided passes: each block processes RB
 // consecutive residues r, so global accesses are RB*32-byte contiguous
 // chunks (the 1-residue variant gathers single 32-B elements at stride
@@ -260,14 +300,27 @@ inline hipError_t ntt_run(Fd<FpCfg>* d_a, Fd<FpCfg>* d_tmp, const NttPlan& plan,
                           ProfFn&& prof) {
   const Fd<FpCfg>* tw = inverse ? plan.d_tw_inv : plan.d_tw_fwd;
   u64 n = 1ULL << k;
-  {
+  constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB), first pass only
+  constexpr int RB = 8;    // residues per block in the strided passes (256-B gathers)
+  // bitrev fused into the first pass's load when that pass exists
+  // (TG_NTT_FUSE_BR=0 restores the separate sweep for A/B)
+  static int fuse_br = [] {
+    const char* e = getenv("TG_NTT_FUSE_BR");
+    return e ? (atoi(e) ? 1 : 0) : 1;
+  }();
+  int s = 1;
+  if (fuse_br && k >= FUSE) {
+    [[maybe_unused]] auto sc = prof(1);
+    u64 ntiles = n >> FUSE;
+    hipLaunchKernelGGL((k_ntt_fused_br<FpCfg, FUSE>),
+                       dim3(ntiles > 2048 ? 2048 : (unsigned)ntiles), dim3(256), 0,
+                       stream, d_tmp, d_a, tw, k);
+    s = 1 + FUSE;
+  } else {
     [[maybe_unused]] auto sc = prof(0);
     hipLaunchKernelGGL(k_bitrev_load<FpCfg>, dim3(ntt_grid(n)), dim3(256), 0, stream,
                        d_tmp, d_a, k, 0, nullptr);
   }
-  constexpr int FUSE = 9;  // 512-element LDS tiles (16 KiB), first pass only
-  constexpr int RB = 8;    // residues per block in the strided passes (256-B gathers)
-  int s = 1;
   while (s <= k) {
     int remaining = k - s + 1;
     if (s == 1 && remaining >= FUSE) {
