@@ -28,5 +28,6 @@ for seed in seeds:
     gp = g.create_proof_raw(inst, adv, RNG)
     if gp != out.raw[:n]:
         bad.append((seed, "mismatch", meta))
+    print(f"seed {seed}: {'MISMATCH' if gp != out.raw[:n] else 'ok'}", flush=True)
 print(f"swept {seeds[0]}..{seed}, bad: {bad if bad else 'none'}")
 g.close()
