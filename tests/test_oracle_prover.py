@@ -74,3 +74,21 @@ def test_mutation_rejection_sample():
         bad = bytearray(proof)
         bad[pos] ^= bit
         assert lib_.orc_verify_cs1(inst, bytes(bad), n) != 0, f"accepted flip at {pos}"
+
+
+def test_gen_bases_pin():
+    """pin the synthetic-base derivation (splitmix64-derived 256-bit
+    multiples of G — the definition both tg_gen_bases and orc_gen_bases
+    implement; changing either silently would un-anchor
+    test_gpu_parity.py::test_gen_bases_parity)."""
+    import hashlib
+    import sys
+
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import oracle_ct as oc
+
+    b = oc.gen_bases(2, 42)
+    assert b[:32].hex().startswith("346674743c1e7b5e")
+    assert b[64:96].hex().startswith("0ef21e4edd2c8ef5")
+    assert (hashlib.blake2b(oc.gen_bases(1024, 42), digest_size=16).hexdigest()
+            == "3b1f0d741a4e60333186874dbcd6d984")
